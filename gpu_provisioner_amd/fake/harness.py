@@ -1,0 +1,142 @@
+"""Full-stack in-process harness: apiserver + fake AKS + real controllers.
+
+Wires the real instance provider, cloudprovider adapter (metrics-decorated),
+and controllers against the in-memory apiserver and the AKS simulator —
+BASELINE.json config #1 ("Single NodeClaim reconciled against fake
+cloudprovider, CPU-only plumbing") and the substrate for bench.py. The
+reference's analogue is its envtest + gomock seams (pkg/fake/), but here the
+whole provision path runs e2e in one process.
+"""
+from __future__ import annotations
+
+import asyncio
+from typing import Optional
+
+from ..apis import v1 as karpv1
+from ..cloudprovider.azure import AzureCloudProvider
+from ..cloudprovider.decorator import MetricsDecorator
+from ..events.recorder import EventRecorder
+from ..kube.informer import Informer, InformerFactory
+from ..kube import objects as ko
+from ..providers.instance.provider import InstanceProvider
+from ..providers.instancetype.catalog import InstanceTypeProvider
+from .agentpools import AKSSimulator, FakeAgentPools
+from .apiserver import InMemoryAPIServer, InMemoryClient
+
+
+class Harness:
+    def __init__(
+        self,
+        *,
+        create_latency: float = 0.0,
+        delete_latency: float = 0.0,
+        api_latency: float = 0.0,
+        ready_latency: float = 0.0,
+        plugin_latency: float = 0.0,
+        node_wait_interval: float = 0.02,
+        region: str = "eastus2",
+    ):
+        self.server = InMemoryAPIServer()
+        self.kube = InMemoryClient(self.server)
+        self.catalog = InstanceTypeProvider(region)
+        self.agent_pools = FakeAgentPools(
+            create_latency=create_latency,
+            delete_latency=delete_latency,
+            api_latency=api_latency,
+        )
+        self.aks = AKSSimulator(
+            self.kube,
+            self.agent_pools,
+            ready_latency=ready_latency,
+            plugin_latency=plugin_latency,
+            gpu_count_for=self.catalog.gpu_count,
+        )
+        self.instances = InstanceProvider(
+            self.agent_pools,
+            self.kube,
+            self.catalog,
+            resource_group="rg",
+            cluster_name="cluster",
+            node_wait_interval=node_wait_interval,
+        )
+        self.cloud = MetricsDecorator(AzureCloudProvider(self.instances, self.catalog))
+        self.recorder = EventRecorder(self.kube)
+        self.informers = InformerFactory(self.kube)
+        self.nodeclaims: Informer = self.informers.informer(karpv1.API_VERSION, karpv1.KIND_NODECLAIM)
+        self.nodes: Informer = self.informers.informer("v1", "Node")
+        self.pods: Informer = self.informers.informer("v1", "Pod")
+        self.pods.add_index("nodeName", lambda o: o.get("spec", {}).get("nodeName") or None)
+        self.controllers: list = []
+        self._started = False
+
+    # -- lifecycle -----------------------------------------------------------
+
+    async def start(self) -> None:
+        self.informers.start_all()
+        await self.informers.wait_for_sync()
+        for c in self.controllers:
+            c.controller.start()
+        self._started = True
+
+    async def stop(self) -> None:
+        for c in self.controllers:
+            await c.controller.stop()
+        await self.informers.stop_all()
+
+    # -- helpers -------------------------------------------------------------
+
+    def make_nodeclaim(
+        self,
+        name: str,
+        vm_size: str = "Standard_ND128isr_MI355X_v6",
+        labels: Optional[dict] = None,
+        **spec_overrides,
+    ) -> dict:
+        nc = karpv1.new_nodeclaim(
+            name,
+            labels=labels if labels is not None else {karpv1.KAITO_WORKSPACE_LABEL_KEY: "ws"},
+        )
+        nc["spec"] = {
+            "requirements": [
+                {
+                    "key": karpv1.INSTANCE_TYPE_LABEL_KEY,
+                    "operator": "In",
+                    "values": [vm_size],
+                }
+            ],
+            "resources": {"requests": {karpv1.AMD_GPU_RESOURCE: str(self.catalog.gpu_count(vm_size) or 8)}},
+            "nodeClassRef": {"group": "kaito.sh", "kind": "KaitoNodeClass", "name": "default"},
+            **spec_overrides,
+        }
+        return nc
+
+    async def wait_for(self, predicate, timeout: float = 10.0, interval: float = 0.01):
+        """Poll an async predicate until truthy; returns its value."""
+        deadline = asyncio.get_event_loop().time() + timeout
+        while True:
+            val = await predicate()
+            if val:
+                return val
+            if asyncio.get_event_loop().time() > deadline:
+                raise TimeoutError("condition not met within timeout")
+            await asyncio.sleep(interval)
+
+    async def wait_initialized(self, name: str, timeout: float = 10.0) -> dict:
+        async def check():
+            try:
+                nc = await self.kube.get(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, name)
+            except Exception:
+                return None
+            return nc if karpv1.is_initialized(nc) else None
+
+        return await self.wait_for(check, timeout)
+
+    async def wait_gone(self, api_version: str, kind: str, name: str, timeout: float = 10.0):
+        async def check():
+            try:
+                await self.kube.get(api_version, kind, name)
+                return None
+            except Exception:
+                return True
+
+        return await self.wait_for(check, timeout)

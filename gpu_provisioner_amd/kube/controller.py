@@ -1,0 +1,152 @@
+"""Controller runner: reconcile workers over a rate-limited queue.
+
+Replaces controller-runtime's controller + the operatorpkg singleton wrapper
+(reference vendor/github.com/awslabs/operatorpkg/singleton/controller.go:22-53
+and vendor/sigs.k8s.io/karpenter/pkg/utils/controller/controller.go's
+CPU-scaled concurrency).
+"""
+from __future__ import annotations
+
+import asyncio
+import logging
+import os
+import time
+from dataclasses import dataclass
+from typing import Awaitable, Callable, Optional
+
+from .workqueue import RateLimiter, RateLimitingQueue
+from ..metrics.registry import (
+    RECONCILE_DURATION,
+    RECONCILE_ERRORS,
+    RECONCILE_TOTAL,
+    WORKQUEUE_DEPTH,
+)
+
+log = logging.getLogger(__name__)
+
+
+@dataclass
+class Result:
+    requeue_after: Optional[float] = None
+    requeue: bool = False
+
+
+def linear_scale_reconciles(minimum: int, maximum: int) -> int:
+    """Scale worker count linearly with available CPUs between min and max —
+    the reference scales lifecycle reconciles 1000-5000 this way
+    (vendor/.../controllers/nodeclaim/lifecycle/controller.go:56-58)."""
+    cpus = os.cpu_count() or 1
+    return max(minimum, min(maximum, minimum * cpus))
+
+
+class Controller:
+    """Runs `reconcile(key)` over keys from a dedup queue with N workers.
+
+    reconcile returns a Result (or None) or raises; errors are logged and
+    retried with per-key exponential backoff.
+    """
+
+    def __init__(
+        self,
+        name: str,
+        reconcile: Callable[[str], Awaitable[Optional[Result]]],
+        workers: int = 1,
+        rate_limiter: Optional[RateLimiter] = None,
+    ):
+        self.name = name
+        self.reconcile = reconcile
+        self.workers = workers
+        self.queue = RateLimitingQueue(rate_limiter, name=name)
+        self._tasks: list = []
+
+    async def enqueue(self, key: str) -> None:
+        await self.queue.add(key)
+
+    async def enqueue_after(self, key: str, delay: float) -> None:
+        await self.queue.add_after(key, delay)
+
+    def enqueue_nowait(self, key: str) -> None:
+        """Synchronous enqueue for informer handlers (not coroutines)."""
+        task = asyncio.get_event_loop().create_task(self.queue.add(key))
+        # keep a reference so the task isn't GC'd before running
+        self._fire_and_forget(task)
+
+    _pending: set = set()
+
+    def _fire_and_forget(self, task: asyncio.Task) -> None:
+        Controller._pending.add(task)
+        task.add_done_callback(Controller._pending.discard)
+
+    def start(self) -> None:
+        for i in range(self.workers):
+            self._tasks.append(
+                asyncio.create_task(self._worker(), name=f"{self.name}-worker-{i}")
+            )
+
+    async def stop(self) -> None:
+        await self.queue.shutdown()
+        for t in self._tasks:
+            t.cancel()
+        for t in self._tasks:
+            try:
+                await t
+            except (asyncio.CancelledError, Exception):
+                pass
+        self._tasks = []
+
+    async def _worker(self) -> None:
+        while True:
+            key = await self.queue.get()
+            if key is None:
+                return
+            WORKQUEUE_DEPTH.labels(controller=self.name).set(self.queue.depth)
+            start = time.monotonic()
+            try:
+                result = await self.reconcile(key)
+            except asyncio.CancelledError:
+                await self.queue.done(key)
+                raise
+            except Exception as e:
+                RECONCILE_ERRORS.labels(controller=self.name).inc()
+                RECONCILE_TOTAL.labels(controller=self.name, result="error").inc()
+                log.warning("%s: reconcile %r failed: %s", self.name, key, e, exc_info=True)
+                await self.queue.done(key)
+                await self.queue.add_rate_limited(key)
+                continue
+            RECONCILE_DURATION.labels(controller=self.name).observe(time.monotonic() - start)
+            RECONCILE_TOTAL.labels(controller=self.name, result="success").inc()
+            self.queue.forget(key)
+            await self.queue.done(key)
+            if result is not None:
+                if result.requeue_after is not None:
+                    await self.queue.add_after(key, result.requeue_after)
+                elif result.requeue:
+                    await self.queue.add_rate_limited(key)
+
+
+class SingletonController(Controller):
+    """A controller that reconciles one synthetic key on a cadence — the shape
+    of both garbage collectors (every 2 min; reference
+    pkg/controllers/instance/garbagecollection/controller.go:123)."""
+
+    KEY = "singleton"
+
+    def __init__(
+        self,
+        name: str,
+        reconcile: Callable[[str], Awaitable[Optional[Result]]],
+        interval: float,
+    ):
+        self.interval = interval
+
+        async def wrapped(key: str) -> Result:
+            result = await reconcile(key)
+            if result is not None and result.requeue_after is not None:
+                return result
+            return Result(requeue_after=self.interval)
+
+        super().__init__(name, wrapped, workers=1)
+
+    def start(self) -> None:
+        super().start()
+        self._fire_and_forget(asyncio.get_event_loop().create_task(self.enqueue(self.KEY)))

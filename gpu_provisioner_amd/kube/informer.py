@@ -1,0 +1,209 @@
+"""Informer: list+watch cache with indexers and event handlers.
+
+Replaces the controller-runtime cache + field indexers the reference registers
+in vendor/sigs.k8s.io/karpenter/pkg/operator/operator.go:250-293 (pod by
+spec.nodeName, node by spec.providerID, nodeclaim by status.providerID,
+volumeattachment by spec.nodeName). Handlers are invoked for every event and
+typically enqueue keys onto a controller's workqueue.
+"""
+from __future__ import annotations
+
+import asyncio
+import logging
+from collections import defaultdict
+from typing import Callable, Optional
+
+from . import objects as ko
+from .client import ADDED, DELETED, GoneError, KubeClient, MODIFIED
+
+log = logging.getLogger(__name__)
+
+
+def object_key(obj: dict) -> str:
+    ns = ko.namespace_of(obj)
+    return f"{ns}/{ko.name_of(obj)}" if ns else ko.name_of(obj)
+
+
+class Informer:
+    """Cache for one (apiVersion, kind), kept fresh by a list+watch loop."""
+
+    def __init__(
+        self,
+        client: KubeClient,
+        api_version: str,
+        kind: str,
+        namespace: str = "",
+        label_selector: str = "",
+        resync_period: float = 0.0,
+    ):
+        self.client = client
+        self.api_version = api_version
+        self.kind = kind
+        self.namespace = namespace
+        self.label_selector = label_selector
+        self.resync_period = resync_period
+        self._cache: dict = {}  # key -> obj
+        self._indexes: dict = {}  # index_name -> (fn, {value: set(keys)})
+        self._handlers: list = []  # fn(event_type, obj)
+        self._synced = asyncio.Event()
+        self._task: Optional[asyncio.Task] = None
+        self._rv = ""
+
+    # -- registration -------------------------------------------------------
+
+    def add_index(self, name: str, fn: Callable) -> None:
+        """fn(obj) -> Optional[str] | list[str]: index value(s) for the object.
+        Idempotent: re-registering an existing index name is a no-op."""
+        if name not in self._indexes:
+            self._indexes[name] = (fn, defaultdict(set))
+
+    def has_index(self, name: str) -> bool:
+        return name in self._indexes
+
+    def add_handler(self, fn: Callable) -> None:
+        """fn(event_type, obj) — called for ADDED/MODIFIED/DELETED."""
+        self._handlers.append(fn)
+
+    # -- cache access -------------------------------------------------------
+
+    def get(self, name: str, namespace: str = "") -> Optional[dict]:
+        key = f"{namespace}/{name}" if namespace else name
+        obj = self._cache.get(key)
+        return ko.deep_copy(obj) if obj else None
+
+    def list(self) -> list:
+        return [ko.deep_copy(o) for o in self._cache.values()]
+
+    def by_index(self, index: str, value: str) -> list:
+        _, idx = self._indexes[index]
+        return [ko.deep_copy(self._cache[k]) for k in idx.get(value, set()) if k in self._cache]
+
+    async def wait_for_sync(self) -> None:
+        await self._synced.wait()
+
+    @property
+    def has_synced(self) -> bool:
+        return self._synced.is_set()
+
+    # -- run loop -----------------------------------------------------------
+
+    def start(self) -> asyncio.Task:
+        if self._task is None:
+            self._task = asyncio.create_task(self._run(), name=f"informer-{self.kind}")
+        return self._task
+
+    async def stop(self) -> None:
+        if self._task:
+            self._task.cancel()
+            try:
+                await self._task
+            except (asyncio.CancelledError, Exception):
+                pass
+            self._task = None
+
+    async def _run(self) -> None:
+        backoff = 0.05
+        while True:
+            try:
+                await self._list_and_watch()
+                backoff = 0.05
+            except asyncio.CancelledError:
+                raise
+            except GoneError:
+                self._rv = ""  # force full relist
+                continue
+            except Exception:
+                log.exception("informer %s list/watch failed; retrying in %.2fs", self.kind, backoff)
+                await asyncio.sleep(backoff)
+                backoff = min(backoff * 2, 5.0)
+
+    async def _list_and_watch(self) -> None:
+        # list
+        if hasattr(self.client, "list_with_rv"):
+            items, rv = await self.client.list_with_rv(  # type: ignore[attr-defined]
+                self.api_version, self.kind, self.namespace, self.label_selector
+            )
+        else:
+            items = await self.client.list(
+                self.api_version, self.kind, self.namespace, self.label_selector
+            )
+            rv = max((int(o["metadata"].get("resourceVersion", 0)) for o in items), default=0)
+            rv = str(rv)
+        self._rv = rv
+        new_keys = {object_key(o) for o in items}
+        # deletions that happened while we weren't watching
+        for key in list(self._cache):
+            if key not in new_keys:
+                gone = self._cache.pop(key)
+                self._reindex(key, gone, remove=True)
+                self._notify(DELETED, gone)
+        for obj in items:
+            self._store(obj, event=MODIFIED if object_key(obj) in self._cache else ADDED)
+        self._synced.set()
+        # watch
+        async for event_type, obj in self.client.watch(
+            self.api_version, self.kind, self.namespace, self._rv, self.label_selector
+        ):
+            self._rv = obj.get("metadata", {}).get("resourceVersion", self._rv)
+            if event_type == DELETED:
+                key = object_key(obj)
+                old = self._cache.pop(key, None)
+                self._reindex(key, old or obj, remove=True)
+                self._notify(DELETED, obj)
+            else:
+                self._store(obj, event=event_type)
+
+    def _store(self, obj: dict, event: str) -> None:
+        key = object_key(obj)
+        old = self._cache.get(key)
+        if old is not None:
+            self._reindex(key, old, remove=True)
+        self._cache[key] = obj
+        self._reindex(key, obj, remove=False)
+        self._notify(event, obj)
+
+    def _reindex(self, key: str, obj: dict, remove: bool) -> None:
+        for fn, idx in self._indexes.values():
+            vals = fn(obj)
+            if vals is None:
+                continue
+            if isinstance(vals, str):
+                vals = [vals]
+            for v in vals:
+                if remove:
+                    idx[v].discard(key)
+                else:
+                    idx[v].add(key)
+
+    def _notify(self, event_type: str, obj: dict) -> None:
+        for h in self._handlers:
+            try:
+                h(event_type, obj)
+            except Exception:
+                log.exception("informer %s handler failed", self.kind)
+
+
+class InformerFactory:
+    """One shared informer per (apiVersion, kind, selector)."""
+
+    def __init__(self, client: KubeClient):
+        self.client = client
+        self._informers: dict = {}
+
+    def informer(
+        self, api_version: str, kind: str, namespace: str = "", label_selector: str = ""
+    ) -> Informer:
+        key = (api_version, kind, namespace, label_selector)
+        if key not in self._informers:
+            self._informers[key] = Informer(self.client, api_version, kind, namespace, label_selector)
+        return self._informers[key]
+
+    def start_all(self) -> list:
+        return [inf.start() for inf in self._informers.values()]
+
+    async def wait_for_sync(self) -> None:
+        await asyncio.gather(*(inf.wait_for_sync() for inf in self._informers.values()))
+
+    async def stop_all(self) -> None:
+        for inf in self._informers.values():
+            await inf.stop()

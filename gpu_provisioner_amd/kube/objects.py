@@ -1,0 +1,379 @@
+"""Kubernetes object model: typed wrappers over the wire-format JSON dicts.
+
+The store/client layer (kube/client.py, fake/apiserver.py) speaks plain JSON
+dicts exactly as the kube-apiserver does; this module provides the typed
+accessors controllers use. Mirrors the subset of apimachinery the reference
+exercises (ObjectMeta, Conditions, OwnerReferences, Taints, resource
+quantities) — see reference vendor/sigs.k8s.io/karpenter/pkg/apis/v1 for the
+behavioral spec.
+"""
+from __future__ import annotations
+
+import copy
+import re
+from dataclasses import dataclass, field
+from datetime import datetime, timezone
+from typing import Any, Iterable, Optional
+
+# ---------------------------------------------------------------------------
+# time helpers (RFC3339, as the apiserver emits)
+# ---------------------------------------------------------------------------
+
+
+def now() -> datetime:
+    return datetime.now(timezone.utc)
+
+
+def fmt_time(t: datetime) -> str:
+    return t.astimezone(timezone.utc).strftime("%Y-%m-%dT%H:%M:%SZ")
+
+
+def parse_time(s: str) -> datetime:
+    # Tolerate fractional seconds and explicit offsets.
+    s = s.strip()
+    if s.endswith("Z"):
+        s = s[:-1] + "+00:00"
+    return datetime.fromisoformat(s)
+
+
+# ---------------------------------------------------------------------------
+# resource.Quantity — the subset of kubernetes quantity arithmetic we need
+# (parsing, comparison, formatting). Supports plain ints/floats, decimal SI
+# suffixes (k, M, G, T, P, m for milli) and binary suffixes (Ki..Pi).
+# ---------------------------------------------------------------------------
+
+_QTY_RE = re.compile(r"^([+-]?[0-9]*\.?[0-9]+)([a-zA-Z]*)$")
+_SUFFIX = {
+    "": 1,
+    "m": 10**-3,
+    "k": 10**3,
+    "M": 10**6,
+    "G": 10**9,
+    "T": 10**12,
+    "P": 10**15,
+    "E": 10**18,
+    "Ki": 2**10,
+    "Mi": 2**20,
+    "Gi": 2**30,
+    "Ti": 2**40,
+    "Pi": 2**50,
+    "Ei": 2**60,
+}
+
+
+class Quantity:
+    """A kubernetes resource quantity. Immutable; compares by numeric value."""
+
+    __slots__ = ("raw", "value")
+
+    def __init__(self, raw: "str | int | float | Quantity"):
+        if isinstance(raw, Quantity):
+            self.raw, self.value = raw.raw, raw.value
+            return
+        if isinstance(raw, (int, float)):
+            self.raw = str(raw)
+            self.value = float(raw)
+            return
+        m = _QTY_RE.match(str(raw))
+        if not m:
+            raise ValueError(f"invalid quantity {raw!r}")
+        num, suf = m.groups()
+        if suf not in _SUFFIX:
+            raise ValueError(f"invalid quantity suffix {suf!r} in {raw!r}")
+        self.raw = str(raw)
+        self.value = float(num) * _SUFFIX[suf]
+
+    def __repr__(self) -> str:
+        return f"Quantity({self.raw!r})"
+
+    def __str__(self) -> str:
+        return self.raw
+
+    def __eq__(self, other: object) -> bool:
+        return isinstance(other, Quantity) and self.value == other.value
+
+    def __lt__(self, other: "Quantity") -> bool:
+        return self.value < other.value
+
+    def __le__(self, other: "Quantity") -> bool:
+        return self.value <= other.value
+
+    def __hash__(self) -> int:
+        return hash(self.value)
+
+    def __add__(self, other: "Quantity") -> "Quantity":
+        return Quantity(_fmt_num(self.value + other.value))
+
+    def __sub__(self, other: "Quantity") -> "Quantity":
+        return Quantity(_fmt_num(self.value - other.value))
+
+    def is_zero(self) -> bool:
+        return self.value == 0
+
+
+def _fmt_num(v: float) -> str:
+    if v == int(v):
+        return str(int(v))
+    return str(v)
+
+
+def qty(v: "str | int | float | Quantity") -> Quantity:
+    return Quantity(v)
+
+
+# ---------------------------------------------------------------------------
+# ObjectMeta-level helpers over plain dict objects
+# ---------------------------------------------------------------------------
+
+
+def meta(obj: dict) -> dict:
+    return obj.setdefault("metadata", {})
+
+
+def name_of(obj: dict) -> str:
+    return obj.get("metadata", {}).get("name", "")
+
+
+def namespace_of(obj: dict) -> str:
+    return obj.get("metadata", {}).get("namespace", "")
+
+
+def uid_of(obj: dict) -> str:
+    return obj.get("metadata", {}).get("uid", "")
+
+
+def labels_of(obj: dict) -> dict:
+    return obj.get("metadata", {}).get("labels") or {}
+
+
+def annotations_of(obj: dict) -> dict:
+    return obj.get("metadata", {}).get("annotations") or {}
+
+
+def set_label(obj: dict, key: str, value: str) -> None:
+    meta(obj).setdefault("labels", {})[key] = value
+
+
+def set_annotation(obj: dict, key: str, value: str) -> None:
+    meta(obj).setdefault("annotations", {})[key] = value
+
+
+def finalizers_of(obj: dict) -> list:
+    return obj.get("metadata", {}).get("finalizers") or []
+
+
+def has_finalizer(obj: dict, fin: str) -> bool:
+    return fin in finalizers_of(obj)
+
+
+def add_finalizer(obj: dict, fin: str) -> bool:
+    fins = meta(obj).setdefault("finalizers", [])
+    if fin in fins:
+        return False
+    fins.append(fin)
+    return True
+
+
+def remove_finalizer(obj: dict, fin: str) -> bool:
+    fins = meta(obj).get("finalizers") or []
+    if fin not in fins:
+        return False
+    fins.remove(fin)
+    return True
+
+
+def deletion_timestamp_of(obj: dict) -> Optional[datetime]:
+    ts = obj.get("metadata", {}).get("deletionTimestamp")
+    return parse_time(ts) if ts else None
+
+
+def creation_timestamp_of(obj: dict) -> Optional[datetime]:
+    ts = obj.get("metadata", {}).get("creationTimestamp")
+    return parse_time(ts) if ts else None
+
+
+def is_deleting(obj: dict) -> bool:
+    return bool(obj.get("metadata", {}).get("deletionTimestamp"))
+
+
+def owner_references_of(obj: dict) -> list:
+    return obj.get("metadata", {}).get("ownerReferences") or []
+
+
+def set_owner_reference(obj: dict, owner: dict, *, block_deletion: bool = True) -> None:
+    """Add/replace an owner reference pointing at `owner` (a full object dict)."""
+    ref = {
+        "apiVersion": owner.get("apiVersion", ""),
+        "kind": owner.get("kind", ""),
+        "name": name_of(owner),
+        "uid": uid_of(owner),
+        "blockOwnerDeletion": block_deletion,
+    }
+    refs = [r for r in owner_references_of(obj) if r.get("uid") != ref["uid"]]
+    refs.append(ref)
+    meta(obj)["ownerReferences"] = refs
+
+
+# ---------------------------------------------------------------------------
+# Conditions (metav1.Condition semantics: type/status/reason/message/
+# lastTransitionTime/observedGeneration)
+# ---------------------------------------------------------------------------
+
+CONDITION_TRUE = "True"
+CONDITION_FALSE = "False"
+CONDITION_UNKNOWN = "Unknown"
+
+
+def get_condition(obj: dict, cond_type: str) -> Optional[dict]:
+    for c in obj.get("status", {}).get("conditions") or []:
+        if c.get("type") == cond_type:
+            return c
+    return None
+
+
+def condition_is(obj: dict, cond_type: str, status: str) -> bool:
+    c = get_condition(obj, cond_type)
+    return bool(c) and c.get("status") == status
+
+
+def condition_is_true(obj: dict, cond_type: str) -> bool:
+    return condition_is(obj, cond_type, CONDITION_TRUE)
+
+
+def set_condition(
+    obj: dict,
+    cond_type: str,
+    status: str,
+    reason: str = "",
+    message: str = "",
+    *,
+    at: Optional[datetime] = None,
+) -> bool:
+    """Set a condition; updates lastTransitionTime only on status change.
+
+    Returns True if anything changed.
+    """
+    conds = obj.setdefault("status", {}).setdefault("conditions", [])
+    existing = None
+    for c in conds:
+        if c.get("type") == cond_type:
+            existing = c
+            break
+    ts = fmt_time(at or now())
+    if existing is None:
+        conds.append(
+            {
+                "type": cond_type,
+                "status": status,
+                "reason": reason or status,
+                "message": message,
+                "lastTransitionTime": ts,
+                "observedGeneration": obj.get("metadata", {}).get("generation", 0),
+            }
+        )
+        return True
+    changed = False
+    if existing.get("status") != status:
+        existing["status"] = status
+        existing["lastTransitionTime"] = ts
+        changed = True
+    if reason and existing.get("reason") != reason:
+        existing["reason"] = reason
+        changed = True
+    if existing.get("message") != message:
+        existing["message"] = message
+        changed = True
+    existing["observedGeneration"] = obj.get("metadata", {}).get("generation", 0)
+    return changed
+
+
+# ---------------------------------------------------------------------------
+# Node helpers (corev1.Node subset)
+# ---------------------------------------------------------------------------
+
+
+def node_is_ready(node: dict) -> bool:
+    for c in node.get("status", {}).get("conditions") or []:
+        if c.get("type") == "Ready":
+            return c.get("status") == CONDITION_TRUE
+    return False
+
+
+def node_ready_condition(node: dict) -> Optional[dict]:
+    for c in node.get("status", {}).get("conditions") or []:
+        if c.get("type") == "Ready":
+            return c
+    return None
+
+
+def node_taints(node: dict) -> list:
+    return node.get("spec", {}).get("taints") or []
+
+
+def node_allocatable(node: dict) -> dict:
+    return node.get("status", {}).get("allocatable") or {}
+
+
+def node_capacity(node: dict) -> dict:
+    return node.get("status", {}).get("capacity") or {}
+
+
+def provider_id_of(node: dict) -> str:
+    return node.get("spec", {}).get("providerID", "")
+
+
+# ---------------------------------------------------------------------------
+# Taints
+# ---------------------------------------------------------------------------
+
+
+@dataclass(frozen=True)
+class Taint:
+    key: str
+    effect: str
+    value: str = ""
+
+    @classmethod
+    def from_dict(cls, d: dict) -> "Taint":
+        return cls(key=d.get("key", ""), effect=d.get("effect", ""), value=d.get("value", ""))
+
+    def to_dict(self) -> dict:
+        d = {"key": self.key, "effect": self.effect}
+        if self.value:
+            d["value"] = self.value
+        return d
+
+    def matches(self, other: "Taint") -> bool:
+        return self.key == other.key and self.effect == other.effect
+
+
+def merge_taints(existing: Iterable[dict], desired: Iterable[dict]) -> list:
+    """Merge `desired` taints into `existing` (key+effect identity), keeping
+    existing entries and appending missing desired ones — the semantics of
+    karpenter's scheduling.Taints.Merge (reference vendor/.../pkg/scheduling/taints.go).
+    """
+    out = [dict(t) for t in existing]
+    have = {(t.get("key"), t.get("effect")) for t in out}
+    for t in desired:
+        if (t.get("key"), t.get("effect")) not in have:
+            out.append(dict(t))
+            have.add((t.get("key"), t.get("effect")))
+    return out
+
+
+# ---------------------------------------------------------------------------
+# misc
+# ---------------------------------------------------------------------------
+
+
+def deep_copy(obj: dict) -> dict:
+    return copy.deepcopy(obj)
+
+
+def group_version_kind(obj: dict) -> tuple:
+    api = obj.get("apiVersion", "")
+    if "/" in api:
+        group, version = api.split("/", 1)
+    else:
+        group, version = "", api
+    return group, version, obj.get("kind", "")

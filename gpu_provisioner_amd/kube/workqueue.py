@@ -1,0 +1,171 @@
+"""Rate-limited work queue with client-go semantics, asyncio-native.
+
+Replaces k8s.io/client-go/util/workqueue (used by every controller in the
+reference via controller-runtime): deduplication (an item is queued at most
+once; re-adds during processing re-queue after Done), delayed adds, per-item
+exponential backoff plus a global token bucket, and retry accounting.
+"""
+from __future__ import annotations
+
+import asyncio
+import heapq
+import time
+from typing import Any, Hashable, Optional
+
+
+class ExponentialBackoff:
+    """Per-item exponential failure backoff: base * 2^failures, capped."""
+
+    def __init__(self, base: float = 0.005, cap: float = 1000.0):
+        self.base = base
+        self.cap = cap
+        self._failures: dict = {}
+
+    def when(self, item: Hashable) -> float:
+        n = self._failures.get(item, 0)
+        self._failures[item] = n + 1
+        return min(self.base * (2**n), self.cap)
+
+    def forget(self, item: Hashable) -> None:
+        self._failures.pop(item, None)
+
+    def num_requeues(self, item: Hashable) -> int:
+        return self._failures.get(item, 0)
+
+
+class TokenBucket:
+    """Global qps/burst limiter — returns the delay an add must wait."""
+
+    def __init__(self, qps: float = 10.0, burst: int = 100):
+        self.qps = qps
+        self.burst = burst
+        self._tokens = float(burst)
+        self._last = time.monotonic()
+
+    def reserve(self) -> float:
+        nw = time.monotonic()
+        self._tokens = min(self.burst, self._tokens + (nw - self._last) * self.qps)
+        self._last = nw
+        self._tokens -= 1.0
+        if self._tokens >= 0:
+            return 0.0
+        return -self._tokens / self.qps
+
+
+class RateLimiter:
+    """Max of exponential per-item backoff and the global bucket (client-go's
+    DefaultControllerRateLimiter shape)."""
+
+    def __init__(self, base: float = 0.005, cap: float = 1000.0, qps: float = 10.0, burst: int = 100):
+        self.backoff = ExponentialBackoff(base, cap)
+        self.bucket = TokenBucket(qps, burst)
+
+    def when(self, item: Hashable) -> float:
+        return max(self.backoff.when(item), self.bucket.reserve())
+
+    def forget(self, item: Hashable) -> None:
+        self.backoff.forget(item)
+
+    def num_requeues(self, item: Hashable) -> int:
+        return self.backoff.num_requeues(item)
+
+
+class RateLimitingQueue:
+    """Async dedup queue with delayed and rate-limited adds."""
+
+    def __init__(self, rate_limiter: Optional[RateLimiter] = None, name: str = ""):
+        self.name = name
+        self.rate_limiter = rate_limiter or RateLimiter()
+        self._queue: list = []  # FIFO of ready items
+        self._dirty: set = set()  # queued or needs requeue
+        self._processing: set = set()
+        self._delayed: list = []  # heap of (ready_at, seq, item)
+        self._seq = 0
+        self._cond = asyncio.Condition()
+        self._shutdown = False
+        self.adds = 0  # metric: total adds
+
+    # -- core ---------------------------------------------------------------
+
+    async def add(self, item: Hashable) -> None:
+        async with self._cond:
+            if self._shutdown or item in self._dirty:
+                return
+            self.adds += 1
+            self._dirty.add(item)
+            if item not in self._processing:
+                self._queue.append(item)
+                self._cond.notify()
+
+    async def add_after(self, item: Hashable, delay: float) -> None:
+        if delay <= 0:
+            await self.add(item)
+            return
+        async with self._cond:
+            if self._shutdown:
+                return
+            self._seq += 1
+            heapq.heappush(self._delayed, (time.monotonic() + delay, self._seq, item))
+            self._cond.notify()
+
+    async def add_rate_limited(self, item: Hashable) -> None:
+        await self.add_after(item, self.rate_limiter.when(item))
+
+    async def get(self) -> Any:
+        """Block until an item is ready; marks it processing. Returns None on shutdown."""
+        async with self._cond:
+            while True:
+                self._drain_delayed()
+                if self._queue:
+                    item = self._queue.pop(0)
+                    self._dirty.discard(item)
+                    self._processing.add(item)
+                    return item
+                if self._shutdown:
+                    return None
+                timeout = self._next_delay()
+                try:
+                    await asyncio.wait_for(self._cond.wait(), timeout)
+                except asyncio.TimeoutError:
+                    pass
+
+    async def done(self, item: Hashable) -> None:
+        async with self._cond:
+            self._processing.discard(item)
+            if item in self._dirty:
+                self._queue.append(item)
+                self._cond.notify()
+
+    def forget(self, item: Hashable) -> None:
+        self.rate_limiter.forget(item)
+
+    def num_requeues(self, item: Hashable) -> int:
+        return self.rate_limiter.num_requeues(item)
+
+    async def shutdown(self) -> None:
+        async with self._cond:
+            self._shutdown = True
+            self._cond.notify_all()
+
+    # -- helpers ------------------------------------------------------------
+
+    def _drain_delayed(self) -> None:
+        nw = time.monotonic()
+        while self._delayed and self._delayed[0][0] <= nw:
+            _, _, item = heapq.heappop(self._delayed)
+            if item not in self._dirty:
+                self._dirty.add(item)
+                if item not in self._processing:
+                    self._queue.append(item)
+
+    def _next_delay(self) -> Optional[float]:
+        if not self._delayed:
+            return None
+        return max(0.0, self._delayed[0][0] - time.monotonic())
+
+    def __len__(self) -> int:
+        return len(self._queue)
+
+    @property
+    def depth(self) -> int:
+        return len(self._queue)

@@ -1,0 +1,285 @@
+"""Instance provider: NodeClaim → single-VM AKS agent pool (MI355X).
+
+The Azure-specific core, re-designed from the behavioral spec of reference
+pkg/providers/instance/instance.go (Create :76-151, Get :153-165, List
+:167-176, Delete :178-187, newAgentPoolObject :321-369, node lookup :371-385)
+and armutils.go (create :28-40, delete-with-state-check :42-76). Differences
+by design:
+  * GPU knowledge comes from the MI355X catalog (providers/instancetype)
+    instead of a `Standard_N` prefix match;
+  * agent pools carry the ROCm bootstrap (gpuProfile, kubelet/sysctl, GPU
+    topology node labels — providers/instance/bootstrap.py);
+  * capacity (amd.com/gpu) is returned on the Instance so launch can
+    pre-populate NodeClaim status.capacity before the node exists.
+"""
+from __future__ import annotations
+
+import asyncio
+import logging
+import random
+import re
+import time
+from typing import Optional
+
+from ...apis import v1 as karpv1
+from ...cloudprovider.types import (
+    CreateError,
+    Instance,
+    InsufficientCapacityError,
+    NodeClaimNotFoundError,
+)
+from ...kube import objects as ko
+from ...kube.client import KubeClient
+from ...utils.utils import parse_agent_pool_name_from_id
+from ..instancetype.catalog import InstanceTypeProvider
+from . import bootstrap
+from .armapi import (
+    AgentPoolsAPI,
+    ARMError,
+    is_arm_not_found,
+    is_create_in_progress,
+    pool_labels,
+    pool_name,
+    pool_props,
+    pool_state,
+    pool_vm_size,
+    taint_to_string,
+)
+
+log = logging.getLogger(__name__)
+
+# agent-pool naming constraint (reference instance.go:50,80-84)
+AGENT_POOL_NAME_RE = re.compile(r"^[a-z][a-z0-9]{0,11}$")
+
+# GC bookkeeping label persisted on the agent pool + node so leak adoption
+# survives controller restarts (reference instance.go:340-342)
+CREATION_TIMESTAMP_LABEL = "kaito.sh/creation-timestamp"
+
+NODE_WAIT_ATTEMPTS = 30
+NODE_WAIT_INTERVAL = 1.0
+NODE_WAIT_JITTER = 0.1
+
+_ARM_THROTTLE_CODES = {"TooManyRequests", "SubscriptionRequestsThrottled"}
+_CAPACITY_CODES = {
+    "SkuNotAvailable",
+    "AllocationFailed",
+    "OverconstrainedAllocationRequest",
+    "ZonalAllocationFailed",
+    "QuotaExceeded",
+    "OperationNotAllowed.QuotaExceeded",
+}
+
+
+class InstanceProvider:
+    def __init__(
+        self,
+        agent_pools: AgentPoolsAPI,
+        kube: KubeClient,
+        catalog: InstanceTypeProvider,
+        resource_group: str,
+        cluster_name: str,
+        *,
+        node_wait_attempts: int = NODE_WAIT_ATTEMPTS,
+        node_wait_interval: float = NODE_WAIT_INTERVAL,
+    ):
+        self.agent_pools = agent_pools
+        self.kube = kube
+        self.catalog = catalog
+        self.resource_group = resource_group
+        self.cluster_name = cluster_name
+        self.node_wait_attempts = node_wait_attempts
+        self.node_wait_interval = node_wait_interval
+
+    # ------------------------------------------------------------------ create
+
+    async def create(self, nodeclaim: dict) -> Instance:
+        name = ko.name_of(nodeclaim)
+        if not AGENT_POOL_NAME_RE.match(name):
+            raise CreateError(
+                f"agent pool name {name!r} must match {AGENT_POOL_NAME_RE.pattern}",
+                condition_reason="InvalidName",
+            )
+        vm_size = self._pick_vm_size(nodeclaim)
+        pool = self.new_agent_pool_object(nodeclaim, vm_size)
+        try:
+            poller = await self.agent_pools.begin_create_or_update(
+                self.resource_group, self.cluster_name, name, pool
+            )
+            created = await poller.result()
+        except ARMError as e:
+            if is_create_in_progress(e):
+                # an earlier create (before a crash) is still running — adopt it
+                log.info("agent pool %s create already in progress; adopting", name)
+                created = await self.agent_pools.get(self.resource_group, self.cluster_name, name)
+            elif e.code in _CAPACITY_CODES:
+                raise InsufficientCapacityError(f"{vm_size}: {e.message or e.code}") from e
+            else:
+                raise CreateError(f"creating agent pool {name}: {e}") from e
+        provider_id, node = await self._wait_for_node(name)
+        if not provider_id:
+            raise CreateError(
+                f"node for agent pool {name} did not register a providerID within "
+                f"{self.node_wait_attempts * self.node_wait_interval:.0f}s",
+                condition_reason="NodeRegistrationTimeout",
+            )
+        return self._to_instance(created, provider_id=provider_id)
+
+    def _pick_vm_size(self, nodeclaim: dict) -> str:
+        values = karpv1.requirement_values(nodeclaim, karpv1.INSTANCE_TYPE_LABEL_KEY)
+        if not values:
+            raise CreateError(
+                f"NodeClaim {ko.name_of(nodeclaim)} has no "
+                f"{karpv1.INSTANCE_TYPE_LABEL_KEY} requirement",
+                condition_reason="NoInstanceType",
+            )
+        # prefer the cheapest catalog SKU among the allowed values; fall back
+        # to the first value for SKUs outside the catalog
+        known = [(v, self.catalog.get(v)) for v in values]
+        priced = [
+            (it.cheapest_offering().price if it.cheapest_offering() else float("inf"), v)
+            for v, it in known
+            if it is not None
+        ]
+        if priced:
+            return min(priced)[1]
+        return values[0]
+
+    def new_agent_pool_object(self, nodeclaim: dict, vm_size: str) -> dict:
+        """Reference newAgentPoolObject (instance.go:321-369), MI355X-native."""
+        labels = dict(ko.labels_of(nodeclaim))
+        labels[karpv1.NODEPOOL_LABEL_KEY] = karpv1.KAITO_NODEPOOL_NAME
+        labels[CREATION_TIMESTAMP_LABEL] = str(int(time.time()))
+        labels.update(bootstrap.gpu_node_labels(vm_size, self.catalog))
+        taints = [taint_to_string(t) for t in nodeclaim.get("spec", {}).get("taints") or []]
+        capacity_type = (
+            karpv1.requirement_values(nodeclaim, karpv1.CAPACITY_TYPE_LABEL_KEY)
+            or [karpv1.CAPACITY_TYPE_ON_DEMAND]
+        )[0]
+        props: dict = {
+            "count": 1,
+            "vmSize": vm_size,
+            "osType": "Linux",
+            "osSKU": bootstrap.determine_os_sku(
+                ko.annotations_of(nodeclaim).get(karpv1.NODE_IMAGE_FAMILY_ANNOTATION_KEY, "")
+            ),
+            "mode": "User",
+            "nodeLabels": labels,
+            "nodeTaints": taints,
+            "tags": {"managed-by": "gpu-provisioner-amd"},
+        }
+        if capacity_type == karpv1.CAPACITY_TYPE_SPOT:
+            props["scaleSetPriority"] = "Spot"
+            props["scaleSetEvictionPolicy"] = "Delete"
+        disk = (
+            nodeclaim.get("spec", {})
+            .get("resources", {})
+            .get("requests", {})
+            .get("ephemeral-storage")
+        )
+        if disk:
+            props["osDiskSizeGB"] = max(1, int(ko.qty(disk).value / 2**30))
+        if self.catalog.is_gpu_sku(vm_size):
+            gpus = self.catalog.gpu_count(vm_size)
+            props["gpuProfile"] = bootstrap.rocm_gpu_profile()
+            props["kubeletConfig"] = bootstrap.rocm_kubelet_config()
+            props["linuxOSConfig"] = bootstrap.rocm_linux_os_config(gpus)
+        return {"name": ko.name_of(nodeclaim), "properties": props}
+
+    async def _wait_for_node(self, pool: str) -> tuple:
+        """Poll for the Node object + providerID (reference instance.go:123-149,
+        getNodesByName :371-385 — lookup via the two agentpool labels)."""
+        for _ in range(self.node_wait_attempts):
+            node = await self._node_for_pool(pool)
+            if node is not None:
+                pid = ko.provider_id_of(node)
+                if pid:
+                    return pid, node
+            await asyncio.sleep(self.node_wait_interval * (1 + random.uniform(0, NODE_WAIT_JITTER)))
+        return "", None
+
+    async def _node_for_pool(self, pool: str) -> Optional[dict]:
+        for selector in (
+            f"{karpv1.AGENTPOOL_LABEL_KEY}={pool}",
+            f"{karpv1.AZURE_AGENTPOOL_LABEL_KEY}={pool}",
+        ):
+            nodes = await self.kube.list("v1", "Node", label_selector=selector)
+            if nodes:
+                return nodes[0]
+        return None
+
+    # --------------------------------------------------------------------- get
+
+    async def get(self, provider_id: str) -> Instance:
+        pool = parse_agent_pool_name_from_id(provider_id)
+        if not pool:
+            raise NodeClaimNotFoundError(f"cannot parse agent pool from providerID {provider_id!r}")
+        try:
+            ap = await self.agent_pools.get(self.resource_group, self.cluster_name, pool)
+        except ARMError as e:
+            if is_arm_not_found(e):
+                raise NodeClaimNotFoundError(f"agent pool {pool} not found") from e
+            raise
+        return self._to_instance(ap, provider_id=provider_id)
+
+    # -------------------------------------------------------------------- list
+
+    async def list(self) -> list:
+        """Agent pools that are kaito-owned and nodeclaim-created (reference
+        instance.go:167-176, predicates :387-413)."""
+        out = []
+        async for ap in self.agent_pools.list(self.resource_group, self.cluster_name):
+            labels = pool_labels(ap)
+            if labels.get(karpv1.NODEPOOL_LABEL_KEY) != karpv1.KAITO_NODEPOOL_NAME:
+                continue
+            if CREATION_TIMESTAMP_LABEL not in labels:
+                continue
+            inst = self._to_instance(ap)
+            node = await self._node_for_pool(pool_name(ap))
+            if node is not None:
+                inst.id = ko.provider_id_of(node)
+            out.append(inst)
+        return out
+
+    # ------------------------------------------------------------------ delete
+
+    async def delete(self, pool: str) -> None:
+        """Skip if already Deleting; NotFound → NodeClaimNotFoundError
+        (reference instance.go:178-187 + armutils.go:42-76)."""
+        try:
+            ap = await self.agent_pools.get(self.resource_group, self.cluster_name, pool)
+        except ARMError as e:
+            if is_arm_not_found(e):
+                raise NodeClaimNotFoundError(f"agent pool {pool} not found") from e
+            raise
+        if pool_state(ap) == "Deleting":
+            return
+        try:
+            poller = await self.agent_pools.begin_delete(
+                self.resource_group, self.cluster_name, pool
+            )
+            await poller.result()
+        except ARMError as e:
+            if is_arm_not_found(e):
+                raise NodeClaimNotFoundError(f"agent pool {pool} not found") from e
+            raise
+
+    # --------------------------------------------------------------- conversion
+
+    def _to_instance(self, ap: dict, provider_id: str = "") -> Instance:
+        props = ap.get("properties", {})
+        labels = pool_labels(ap)
+        return Instance(
+            name=pool_name(ap),
+            id=provider_id,
+            type=pool_vm_size(ap),
+            state=pool_state(ap),
+            image_id=props.get("nodeImageVersion", ""),
+            capacity_type=(
+                karpv1.CAPACITY_TYPE_SPOT
+                if props.get("scaleSetPriority") == "Spot"
+                else karpv1.CAPACITY_TYPE_ON_DEMAND
+            ),
+            labels=dict(labels),
+            tags=dict(props.get("tags") or {}),
+            created_at=labels.get(CREATION_TIMESTAMP_LABEL, ""),
+        )
