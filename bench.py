@@ -1,0 +1,176 @@
+#!/usr/bin/env python3
+"""Benchmark: NodeClaim provisioning throughput + p50 Ready latency.
+
+Measures BASELINE.json's north-star metric on config #1 ("Single NodeClaim
+reconciled against fake cloudprovider — CPU-only plumbing, no cloud"): the
+full controller stack (in-memory apiserver + informers + lifecycle controller
++ instance provider + fake AKS with zero cloud latency) provisioning and
+tearing down batches of MI355X NodeClaims. One *step* = provision
+`--concurrent` NodeClaims to Initialized (amd.com/gpu registered), then
+delete them to completion (full churn cycle). The reference (Azure/
+gpu-provisioner) publishes no numbers, so vs_baseline is null;
+the implicit envelope to beat is its fixed per-claim overhead (≥1 s
+post-patch sleep on the provision path + 5 s termination polls).
+
+Usage: python bench.py [--gpus N] [--steps K] [--warmup W] [--concurrent C]
+For N>1 the driver launches this under torch.distributed.run, one rank per
+GPU; the workload is controller plumbing (CPU-bound), ranks run independent
+provisioner stacks and aggregate via gloo (weak scaling).
+"""
+from __future__ import annotations
+
+import argparse
+import asyncio
+import json
+import os
+import statistics
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from gpu_provisioner_amd.apis import v1 as karpv1  # noqa: E402
+from gpu_provisioner_amd.controllers.lifecycle.controller import LifecycleController  # noqa: E402
+from gpu_provisioner_amd.fake.harness import Harness  # noqa: E402
+
+VM_SIZE = "Standard_ND128isr_MI355X_v6"
+
+
+def build_harness() -> Harness:
+    h = Harness(node_wait_interval=0.002)
+    h.lifecycle = LifecycleController(
+        h.kube, h.cloud, h.recorder, h.nodeclaims, h.nodes,
+        workers=256, termination_requeue=0.002,
+    )
+    h.controllers.append(h.lifecycle)
+    return h
+
+
+async def one_step(h: Harness, step: int, concurrent: int, latencies: list) -> None:
+    names = [f"s{step % 1000:03d}c{i:02d}" for i in range(concurrent)]
+
+    async def provision(name: str) -> None:
+        t0 = time.monotonic()
+        await h.kube.create(h.make_nodeclaim(name, VM_SIZE))
+        await h.wait_initialized(name, timeout=60.0)
+        latencies.append(time.monotonic() - t0)
+
+    await asyncio.gather(*(provision(n) for n in names))
+
+    async def teardown(name: str) -> None:
+        await h.kube.delete(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, name)
+        await h.wait_gone(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, name, timeout=60.0)
+
+    await asyncio.gather(*(teardown(n) for n in names))
+
+
+async def run_bench(steps: int, warmup: int, concurrent: int) -> dict:
+    h = build_harness()
+    await h.start()
+    try:
+        warm_lat: list = []
+        for s in range(warmup):
+            await one_step(h, s, concurrent, warm_lat)
+        latencies: list = []
+        t0 = time.monotonic()
+        for s in range(warmup, warmup + steps):
+            await one_step(h, s, concurrent, latencies)
+        elapsed = time.monotonic() - t0
+        return {"elapsed_s": elapsed, "latencies": latencies}
+    finally:
+        await h.stop()
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--concurrent", type=int, default=8,
+                    help="NodeClaims provisioned per step (8 = one full MI355X host worth)")
+    args = ap.parse_args()
+
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    dist = None
+    if world_size > 1:
+        import torch.distributed as dist  # type: ignore
+
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        dist.init_process_group(backend="gloo", rank=rank, world_size=world_size)
+
+    try:
+        import torch
+
+        cuda = torch.cuda.is_available()
+    except Exception:
+        torch, cuda = None, False
+
+    def sync():
+        if dist is not None:
+            dist.barrier()
+        if cuda:
+            torch.cuda.synchronize()
+
+    sync()
+    result = asyncio.run(run_bench(args.steps, args.warmup, args.concurrent))
+    sync()
+
+    elapsed = result["elapsed_s"]
+    if dist is not None:
+        import torch as _t
+
+        buf = _t.tensor([elapsed], dtype=_t.float64)
+        dist.all_reduce(buf, op=dist.ReduceOp.MAX)
+        max_elapsed = float(buf.item())
+        all_lat: list = [None] * world_size
+        dist.all_gather_object(all_lat, result["latencies"])
+        latencies = [x for part in all_lat for x in part]
+    else:
+        max_elapsed = elapsed
+        latencies = result["latencies"]
+
+    total_claims = args.steps * args.concurrent * world_size
+    value = total_claims / (max_elapsed / 60.0)
+    p50 = statistics.median(latencies) if latencies else float("nan")
+    p95 = (
+        statistics.quantiles(latencies, n=20)[18]
+        if len(latencies) >= 20
+        else (max(latencies) if latencies else float("nan"))
+    )
+
+    if rank == 0:
+        print(
+            json.dumps(
+                {
+                    "metric": "nodeclaims_per_min",
+                    "value": round(value, 2),
+                    "unit": "NodeClaims/min",
+                    "n_gpus": world_size,
+                    "steps": args.steps,
+                    "warmup": args.warmup,
+                    "ms_per_step": round(max_elapsed / args.steps * 1000.0, 3),
+                    "higher_is_better": True,
+                    "scaling": "weak",
+                    "vs_baseline": None,
+                    "dtype": "n/a",
+                    "data": "synthetic",
+                    "config": {
+                        "model": "MI355X NodeClaim provisioning (karpenter CloudProvider contract)",
+                        "vm_sku": VM_SIZE,
+                        "concurrent_nodeclaims": args.concurrent,
+                        "cloud": "in-memory fake AKS, zero latency (BASELINE config #1)",
+                        "cycle": "create→Launched→Registered→Initialized(amd.com/gpu)→delete→gone",
+                        "p50_ready_latency_s": round(p50, 4),
+                        "p95_ready_latency_s": round(p95, 4),
+                        "parallelism": f"dp{world_size}" if world_size > 1 else "single",
+                    },
+                }
+            )
+        )
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,194 @@
+"""Python binding + CLI for the MI355X node agent (nodeagent/agent.hip).
+
+Runs on provisioned nodes (DaemonSet) to validate the GPU stack before the
+node is trusted, and feeds the node.health repair policy with on-node
+evidence. Fails LOUDLY if the native library is missing on a GPU machine —
+there is no eager/python fallback for the hardware checks.
+
+CLI: ``python -m gpu_provisioner_amd.nodeagent [--json] [--expect-gpus N]``
+Exit 0 = healthy, 1 = unhealthy/degraded, 2 = agent error.
+"""
+from __future__ import annotations
+
+import argparse
+import ctypes
+import json
+import os
+import sys
+from dataclasses import asdict, dataclass, field
+
+_LIB_NAME = "libmi355x_nodeagent.so"
+_SEARCH_PATHS = (
+    os.path.join(os.path.dirname(__file__), "_native", _LIB_NAME),
+    os.path.join(os.path.dirname(__file__), "..", "nodeagent", _LIB_NAME),
+)
+
+EXPECTED_ARCH = "gfx950"
+EXPECTED_HBM_GB = 288
+# healthy-node floors (MI355X: 8 TB/s HBM peak, ~6.3 TB/s achievable float4
+# copy; xGMI 7 links x ~153 GB/s guide values)
+MIN_HBM_BW_GBS = 4000.0
+MIN_XGMI_BW_GBS = 30.0
+
+
+class NodeAgentError(RuntimeError):
+    pass
+
+
+def _load_lib() -> ctypes.CDLL:
+    for p in _SEARCH_PATHS:
+        if os.path.exists(p):
+            return ctypes.CDLL(os.path.abspath(p))
+    raise NodeAgentError(
+        f"{_LIB_NAME} not found (searched {', '.join(_SEARCH_PATHS)}); "
+        "build it with __graft_entry__.build() / make -C nodeagent"
+    )
+
+
+@dataclass
+class GPUReport:
+    index: int
+    name: str = ""
+    arch: str = ""
+    hbm_gb: float = 0.0
+    cus: int = 0
+    hbm_bw_gbs: float = 0.0
+    fma_ok: bool = False
+    mfma_ok: bool = False
+    healthy: bool = False
+    problems: list = field(default_factory=list)
+
+
+@dataclass
+class NodeReport:
+    healthy: bool = False
+    gpu_count: int = 0
+    gpus: list = field(default_factory=list)
+    xgmi_p2p: list = field(default_factory=list)
+    xgmi_bw_gbs: list = field(default_factory=list)
+    problems: list = field(default_factory=list)
+
+
+class NodeAgent:
+    def __init__(self):
+        self.lib = _load_lib()
+        self.lib.na_last_error.restype = ctypes.c_char_p
+
+    def _err(self) -> str:
+        return (self.lib.na_last_error() or b"").decode()
+
+    def device_count(self) -> int:
+        n = ctypes.c_int(0)
+        if self.lib.na_device_count(ctypes.byref(n)) != 0:
+            raise NodeAgentError(f"device_count: {self._err()}")
+        return n.value
+
+    def device_info(self, dev: int) -> tuple:
+        name = ctypes.create_string_buffer(256)
+        arch = ctypes.create_string_buffer(256)
+        hbm = ctypes.c_longlong(0)
+        cus = ctypes.c_int(0)
+        clk = ctypes.c_int(0)
+        rc = self.lib.na_device_info(
+            dev, name, 256, arch, 256, ctypes.byref(hbm), ctypes.byref(cus), ctypes.byref(clk)
+        )
+        if rc != 0:
+            raise NodeAgentError(f"device_info({dev}): {self._err()}")
+        return name.value.decode(), arch.value.decode(), hbm.value, cus.value
+
+    def hbm_bandwidth(self, dev: int, bytes_: int = 1 << 30, iters: int = 10) -> float:
+        out = ctypes.c_double(0)
+        rc = self.lib.na_hbm_bandwidth(
+            dev, ctypes.c_longlong(bytes_), iters, ctypes.byref(out)
+        )
+        if rc != 0:
+            raise NodeAgentError(f"hbm_bandwidth({dev}): {self._err()}")
+        return out.value
+
+    def fma_selftest(self, dev: int) -> bool:
+        return self.lib.na_fma_selftest(dev) == 0
+
+    def mfma_selftest(self, dev: int) -> bool:
+        return self.lib.na_mfma_selftest(dev) == 0
+
+    def p2p_matrix(self, n: int) -> list:
+        buf = (ctypes.c_int * (n * n))()
+        if self.lib.na_p2p_matrix(n, buf) != 0:
+            raise NodeAgentError(f"p2p_matrix: {self._err()}")
+        return [[buf[i * n + j] for j in range(n)] for i in range(n)]
+
+    def p2p_bandwidth(self, src: int, dst: int, bytes_: int = 256 << 20, iters: int = 5) -> float:
+        out = ctypes.c_double(0)
+        rc = self.lib.na_p2p_bandwidth(
+            src, dst, ctypes.c_longlong(bytes_), iters, ctypes.byref(out)
+        )
+        if rc != 0:
+            raise NodeAgentError(f"p2p_bandwidth({src},{dst}): {self._err()}")
+        return out.value
+
+    # -- full health check ---------------------------------------------------
+
+    def check(self, expect_gpus: int = 0, bw_bytes: int = 1 << 30) -> NodeReport:
+        report = NodeReport()
+        report.gpu_count = self.device_count()
+        if expect_gpus and report.gpu_count != expect_gpus:
+            report.problems.append(
+                f"expected {expect_gpus} GPUs, found {report.gpu_count}"
+            )
+        for d in range(report.gpu_count):
+            g = GPUReport(index=d)
+            try:
+                g.name, g.arch, hbm, g.cus = self.device_info(d)
+                g.hbm_gb = round(hbm / 1e9, 1)
+                if EXPECTED_ARCH not in g.arch:
+                    g.problems.append(f"arch {g.arch} != {EXPECTED_ARCH}")
+                g.hbm_bw_gbs = round(self.hbm_bandwidth(d, bw_bytes), 1)
+                if g.hbm_bw_gbs < MIN_HBM_BW_GBS:
+                    g.problems.append(
+                        f"HBM bandwidth {g.hbm_bw_gbs} GB/s below floor {MIN_HBM_BW_GBS}"
+                    )
+                g.fma_ok = self.fma_selftest(d)
+                if not g.fma_ok:
+                    g.problems.append(f"VALU FMA selftest failed: {self._err()}")
+                g.mfma_ok = self.mfma_selftest(d)
+                if not g.mfma_ok:
+                    g.problems.append(f"MFMA matrix-pipe selftest failed: {self._err()}")
+            except NodeAgentError as e:
+                g.problems.append(str(e))
+            g.healthy = not g.problems
+            report.gpus.append(g)
+        if report.gpu_count > 1:
+            report.xgmi_p2p = self.p2p_matrix(report.gpu_count)
+            for i in range(report.gpu_count):
+                for j in range(report.gpu_count):
+                    if i != j and not report.xgmi_p2p[i][j]:
+                        report.problems.append(f"no xGMI peer access {i}->{j}")
+        report.problems.extend(
+            p for g in report.gpus for p in (f"gpu{g.index}: {q}" for q in g.problems)
+        )
+        report.healthy = not report.problems and report.gpu_count > 0
+        return report
+
+
+def main(argv=None) -> int:
+    ap = argparse.ArgumentParser(description="MI355X node health agent")
+    ap.add_argument("--json", action="store_true", help="emit JSON report")
+    ap.add_argument("--expect-gpus", type=int, default=0)
+    ap.add_argument("--bw-bytes", type=int, default=1 << 30)
+    args = ap.parse_args(argv)
+    try:
+        agent = NodeAgent()
+        report = agent.check(expect_gpus=args.expect_gpus, bw_bytes=args.bw_bytes)
+    except NodeAgentError as e:
+        print(json.dumps({"healthy": False, "agent_error": str(e)}))
+        return 2
+    out = asdict(report)
+    if args.json:
+        print(json.dumps(out, indent=2))
+    else:
+        print(json.dumps(out))
+    return 0 if report.healthy else 1
+
+
+if __name__ == "__main__":
+    sys.exit(main())
