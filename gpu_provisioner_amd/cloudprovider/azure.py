@@ -26,6 +26,10 @@ class AzureCloudProvider(CloudProvider):
     def __init__(self, instances: InstanceProvider, catalog: InstanceTypeProvider):
         self.instances = instances
         self.catalog = catalog
+        self._repair_policies = [
+            RepairPolicy("Ready", ko.CONDITION_FALSE, NODE_REPAIR_TOLERATION_SECONDS),
+            RepairPolicy("Ready", ko.CONDITION_UNKNOWN, NODE_REPAIR_TOLERATION_SECONDS),
+        ]
 
     async def create(self, nodeclaim: dict) -> dict:
         instance = await self.instances.create(nodeclaim)
@@ -50,10 +54,7 @@ class AzureCloudProvider(CloudProvider):
         return ""
 
     def repair_policies(self) -> list:
-        return [
-            RepairPolicy("Ready", ko.CONDITION_FALSE, NODE_REPAIR_TOLERATION_SECONDS),
-            RepairPolicy("Ready", ko.CONDITION_UNKNOWN, NODE_REPAIR_TOLERATION_SECONDS),
-        ]
+        return self._repair_policies
 
     def name(self) -> str:
         return "azure"

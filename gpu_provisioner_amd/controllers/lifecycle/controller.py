@@ -233,9 +233,16 @@ class LifecycleController:
         node_taints = ko.merge_taints(
             ko.node_taints(node), nodeclaim.get("spec", {}).get("taints") or []
         )
+        node_finalizers = list(ko.finalizers_of(node))
+        if karpv1.TERMINATION_FINALIZER not in node_finalizers:
+            # the node carries the termination finalizer so node deletion runs
+            # the drain pipeline (node.termination controller) before the
+            # kubelet object vanishes
+            node_finalizers.append(karpv1.TERMINATION_FINALIZER)
         patch: dict = {
             "metadata": {
                 "labels": node_labels,
+                "finalizers": node_finalizers,
                 "ownerReferences": [
                     {
                         "apiVersion": karpv1.API_VERSION,

@@ -66,8 +66,61 @@ class Harness:
         self.nodes: Informer = self.informers.informer("v1", "Node")
         self.pods: Informer = self.informers.informer("v1", "Pod")
         self.pods.add_index("nodeName", lambda o: o.get("spec", {}).get("nodeName") or None)
+        self.volumeattachments: Informer = self.informers.informer(
+            "storage.k8s.io/v1", "VolumeAttachment"
+        )
         self.controllers: list = []
         self._started = False
+
+    def add_all_controllers(
+        self,
+        *,
+        lifecycle_workers: int = 64,
+        termination_requeue: float = 0.05,
+        drain_requeue: float = 0.05,
+        instance_poll: float = 0.05,
+        gc_interval: float = 0.5,
+        adoption_age: float = 0.2,
+        with_health: bool = True,
+    ) -> "Harness":
+        """Wire the full controller set (the production main() topology) with
+        test-friendly cadences."""
+        from ..controllers.garbagecollection.controller import (
+            InstanceGCController,
+            NodeClaimGCController,
+        )
+        from ..controllers.health.controller import HealthController
+        from ..controllers.lifecycle.controller import LifecycleController
+        from ..controllers.termination.controller import TerminationController
+        from ..controllers.termination.eviction import EvictionQueue
+
+        self.eviction_queue = EvictionQueue(self.kube, self.recorder, workers=8)
+        self.lifecycle = LifecycleController(
+            self.kube, self.cloud, self.recorder, self.nodeclaims, self.nodes,
+            workers=lifecycle_workers, termination_requeue=termination_requeue,
+        )
+        self.termination = TerminationController(
+            self.kube, self.cloud, self.recorder, self.nodes, self.nodeclaims,
+            self.pods, self.volumeattachments, self.eviction_queue,
+            workers=32, drain_requeue=drain_requeue, instance_poll=instance_poll,
+        )
+        self.instance_gc = InstanceGCController(
+            self.kube, self.cloud, self.recorder,
+            interval=gc_interval, adoption_age=adoption_age,
+        )
+        self.nodeclaim_gc = NodeClaimGCController(
+            self.kube, self.cloud, self.recorder, interval=gc_interval
+        )
+        self.controllers += [
+            self.eviction_queue, self.lifecycle, self.termination,
+            self.instance_gc, self.nodeclaim_gc,
+        ]
+        if with_health:
+            self.health = HealthController(
+                self.kube, self.cloud, self.recorder, self.nodes
+            )
+            self.controllers.append(self.health)
+        return self
 
     # -- lifecycle -----------------------------------------------------------
 

@@ -148,7 +148,7 @@ class InstanceProvider:
         """Reference newAgentPoolObject (instance.go:321-369), MI355X-native."""
         labels = dict(ko.labels_of(nodeclaim))
         labels[karpv1.NODEPOOL_LABEL_KEY] = karpv1.KAITO_NODEPOOL_NAME
-        labels[CREATION_TIMESTAMP_LABEL] = str(int(time.time()))
+        labels[CREATION_TIMESTAMP_LABEL] = str(int(time.time() * 1000))  # unix ms
         labels.update(bootstrap.gpu_node_labels(vm_size, self.catalog))
         taints = [taint_to_string(t) for t in nodeclaim.get("spec", {}).get("taints") or []]
         capacity_type = (
