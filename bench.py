@@ -30,20 +30,22 @@ import time
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
 from gpu_provisioner_amd.apis import v1 as karpv1  # noqa: E402
-from gpu_provisioner_amd.controllers.lifecycle.controller import LifecycleController  # noqa: E402
 from gpu_provisioner_amd.fake.harness import Harness  # noqa: E402
 
 VM_SIZE = "Standard_ND128isr_MI355X_v6"
 
 
 def build_harness() -> Harness:
-    h = Harness(node_wait_interval=0.002)
-    h.lifecycle = LifecycleController(
-        h.kube, h.cloud, h.recorder, h.nodeclaims, h.nodes,
-        workers=256, termination_requeue=0.002,
+    # the full production controller topology: lifecycle + termination +
+    # eviction + both GCs (health excluded: no repairs during a clean bench)
+    return Harness(node_wait_interval=0.002).add_all_controllers(
+        lifecycle_workers=256,
+        termination_requeue=0.002,
+        drain_requeue=0.002,
+        instance_poll=0.002,
+        gc_interval=30.0,
+        with_health=False,
     )
-    h.controllers.append(h.lifecycle)
-    return h
 
 
 async def one_step(h: Harness, step: int, concurrent: int, latencies: list) -> None:

@@ -17,13 +17,11 @@ from tests.conftest import run
 
 
 def make_harness(**kw) -> Harness:
-    h = Harness(**kw)
-    h.lifecycle = LifecycleController(
-        h.kube, h.cloud, h.recorder, h.nodeclaims, h.nodes, workers=16,
-        termination_requeue=0.05,
+    # lifecycle always runs with termination + eviction (registration puts the
+    # termination finalizer on Nodes; the termination controller removes it)
+    return Harness(**kw).add_all_controllers(
+        lifecycle_workers=16, termination_requeue=0.05, with_health=False
     )
-    h.controllers.append(h.lifecycle)
-    return h
 
 
 def test_provision_to_initialized_full_path():
