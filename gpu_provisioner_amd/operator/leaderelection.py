@@ -1,0 +1,120 @@
+"""Lease-based leader election (coordination.k8s.io/v1).
+
+Replaces controller-runtime's leaderelection (reference wires it at
+vendor/.../pkg/operator/operator.go:157-163, disabled by default per
+options.go:117): acquire a Lease, renew on an interval, yield leadership on
+renewal failure. Controllers start only while leading.
+"""
+from __future__ import annotations
+
+import asyncio
+import logging
+import socket
+import uuid
+from typing import Callable, Optional
+
+from ..kube import objects as ko
+from ..kube.client import ConflictError, KubeClient, NotFoundError
+
+log = logging.getLogger(__name__)
+
+LEASE_DURATION = 15.0
+RENEW_INTERVAL = 5.0
+RETRY_INTERVAL = 2.0
+
+
+class LeaderElector:
+    def __init__(
+        self,
+        kube: KubeClient,
+        name: str,
+        namespace: str,
+        identity: str = "",
+        lease_duration: float = LEASE_DURATION,
+        renew_interval: float = RENEW_INTERVAL,
+    ):
+        self.kube = kube
+        self.name = name
+        self.namespace = namespace
+        self.identity = identity or f"{socket.gethostname()}_{uuid.uuid4().hex[:8]}"
+        self.lease_duration = lease_duration
+        self.renew_interval = renew_interval
+        self.is_leader = False
+        self._task: Optional[asyncio.Task] = None
+
+    async def run(self, on_started_leading: Callable, on_stopped_leading: Callable) -> None:
+        """Blocks forever: acquire → lead (callback) → renew loop → on failure
+        release and re-acquire."""
+        while True:
+            acquired = await self._try_acquire()
+            if not acquired:
+                await asyncio.sleep(RETRY_INTERVAL)
+                continue
+            self.is_leader = True
+            log.info("leader election: acquired lease %s/%s", self.namespace, self.name)
+            await on_started_leading()
+            try:
+                while True:
+                    await asyncio.sleep(self.renew_interval)
+                    if not await self._renew():
+                        break
+            finally:
+                self.is_leader = False
+                log.warning("leader election: lost lease %s/%s", self.namespace, self.name)
+                await on_stopped_leading()
+
+    def _lease(self) -> dict:
+        now = ko.fmt_time(ko.now())
+        return {
+            "apiVersion": "coordination.k8s.io/v1",
+            "kind": "Lease",
+            "metadata": {"name": self.name, "namespace": self.namespace},
+            "spec": {
+                "holderIdentity": self.identity,
+                "leaseDurationSeconds": int(self.lease_duration),
+                "acquireTime": now,
+                "renewTime": now,
+            },
+        }
+
+    async def _try_acquire(self) -> bool:
+        try:
+            cur = await self.kube.get("coordination.k8s.io/v1", "Lease", self.name, self.namespace)
+        except NotFoundError:
+            try:
+                await self.kube.create(self._lease())
+                return True
+            except Exception:
+                return False
+        spec = cur.get("spec", {})
+        holder = spec.get("holderIdentity", "")
+        renew = spec.get("renewTime")
+        expired = True
+        if renew:
+            try:
+                age = (ko.now() - ko.parse_time(renew)).total_seconds()
+                expired = age > spec.get("leaseDurationSeconds", self.lease_duration)
+            except ValueError:
+                pass
+        if holder == self.identity or expired or not holder:
+            cur["spec"] = {**spec, **self._lease()["spec"]}
+            try:
+                await self.kube.update(cur)
+                return True
+            except (ConflictError, NotFoundError):
+                return False
+        return False
+
+    async def _renew(self) -> bool:
+        try:
+            cur = await self.kube.get("coordination.k8s.io/v1", "Lease", self.name, self.namespace)
+        except Exception:
+            return False
+        if cur.get("spec", {}).get("holderIdentity") != self.identity:
+            return False
+        cur["spec"]["renewTime"] = ko.fmt_time(ko.now())
+        try:
+            await self.kube.update(cur)
+            return True
+        except Exception:
+            return False

@@ -1,0 +1,156 @@
+"""Manager: the operator runtime — servers, informers, controllers, leases.
+
+Replaces controller-runtime's Manager as the reference wires it
+(vendor/sigs.k8s.io/karpenter/pkg/operator/operator.go:155-248): metrics
+server on :8080 (prometheus + optional profiling endpoints), health probes on
+:8081 (healthz; readyz = informer caches synced + required CRDs present),
+informer start/sync, controller start, optional Lease leader election.
+"""
+from __future__ import annotations
+
+import asyncio
+import logging
+import sys
+import traceback
+from typing import Optional
+
+from prometheus_client import generate_latest, CONTENT_TYPE_LATEST
+from starlette.applications import Starlette
+from starlette.responses import JSONResponse, PlainTextResponse, Response
+from starlette.routing import Route
+import uvicorn
+
+from ..kube.client import KubeClient, NotFoundError
+from ..kube.informer import InformerFactory
+from ..metrics.registry import BUILD_INFO
+from .leaderelection import LeaderElector
+from .options import Options
+
+log = logging.getLogger(__name__)
+
+
+class Manager:
+    def __init__(
+        self,
+        client: KubeClient,
+        options: Options,
+        informers: Optional[InformerFactory] = None,
+        required_crds: tuple = (),
+        version: str = "0.1.0",
+    ):
+        self.client = client
+        self.options = options
+        self.informers = informers or InformerFactory(client)
+        self.required_crds = required_crds  # [(api_version, kind)] probed via list
+        self.version = version
+        self.controllers: list = []
+        self._elector: Optional[LeaderElector] = None
+        self._server_tasks: list = []
+        self._started = asyncio.Event()
+
+    def register(self, *controllers) -> "Manager":
+        self.controllers.extend(controllers)
+        return self
+
+    # ----------------------------------------------------------- http apps
+
+    def _metrics_app(self) -> Starlette:
+        async def metrics(request):
+            return Response(generate_latest(), media_type=CONTENT_TYPE_LATEST)
+
+        routes = [Route("/metrics", metrics)]
+        if self.options.enable_profiling:
+            # pprof-equivalent debug endpoints (reference operator.go:181-197)
+            async def stacks(request):
+                frames = sys._current_frames()
+                out = {}
+                for tid, frame in frames.items():
+                    out[str(tid)] = "".join(traceback.format_stack(frame))
+                return JSONResponse(out)
+
+            async def tasks(request):
+                return JSONResponse(
+                    [repr(t) for t in asyncio.all_tasks()], status_code=200
+                )
+
+            routes += [
+                Route("/debug/pprof/goroutine", stacks),
+                Route("/debug/tasks", tasks),
+            ]
+        return Starlette(routes=routes)
+
+    def _probes_app(self) -> Starlette:
+        async def healthz(request):
+            return PlainTextResponse("ok")
+
+        async def readyz(request):
+            # cache sync check
+            for key, inf in self.informers._informers.items():
+                if not inf.has_synced:
+                    return PlainTextResponse(f"informer {key} not synced", status_code=503)
+            # CRD presence check (reference operator.go:203-221, NodeClaim only)
+            for api_version, kind in self.required_crds:
+                try:
+                    await self.client.list(api_version, kind)
+                except NotFoundError:
+                    return PlainTextResponse(f"CRD {kind} absent", status_code=503)
+                except Exception as e:
+                    return PlainTextResponse(f"CRD check {kind}: {e}", status_code=503)
+            return PlainTextResponse("ok")
+
+        return Starlette(routes=[Route("/healthz", healthz), Route("/readyz", readyz)])
+
+    async def _serve(self, app, port: int) -> None:
+        config = uvicorn.Config(app, host="0.0.0.0", port=port, log_level="warning", lifespan="off")
+        server = uvicorn.Server(config)
+        await server.serve()
+
+    # ------------------------------------------------------------- lifecycle
+
+    async def start(self, serve_http: bool = True) -> None:
+        BUILD_INFO.labels(version=self.version).set(1)
+        if serve_http:
+            self._server_tasks = [
+                asyncio.create_task(
+                    self._serve(self._metrics_app(), self.options.metrics_port),
+                    name="metrics-server",
+                ),
+                asyncio.create_task(
+                    self._serve(self._probes_app(), self.options.health_probe_port),
+                    name="probes-server",
+                ),
+            ]
+        self.informers.start_all()
+        await self.informers.wait_for_sync()
+        if self.options.leader_elect:
+            self._elector = LeaderElector(
+                self.client,
+                self.options.leader_election_name,
+                self.options.leader_election_namespace,
+            )
+            asyncio.create_task(
+                self._elector.run(self._start_controllers, self._stop_controllers),
+                name="leader-elector",
+            )
+        else:
+            await self._start_controllers()
+        self._started.set()
+
+    async def _start_controllers(self) -> None:
+        for c in self.controllers:
+            c.controller.start()
+        log.info("started %d controllers", len(self.controllers))
+
+    async def _stop_controllers(self) -> None:
+        for c in self.controllers:
+            await c.controller.stop()
+
+    async def run_forever(self) -> None:
+        await self.start()
+        await asyncio.Event().wait()
+
+    async def stop(self) -> None:
+        await self._stop_controllers()
+        await self.informers.stop_all()
+        for t in self._server_tasks:
+            t.cancel()
