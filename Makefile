@@ -1,0 +1,81 @@
+# gpu-provisioner-amd — target surface mirroring the reference Makefile
+# (build, unit-test, e2etests, docker-build, az-* setup) adapted to the
+# Python+HIP toolchain.
+
+VERSION ?= 0.1.0
+IMAGE ?= ghcr.io/kaito-project/gpu-provisioner-amd:$(VERSION)
+PYTHON ?= python
+HIPCC ?= hipcc
+OFFLOAD_ARCH ?= gfx950
+
+AZURE_SUBSCRIPTION_ID ?=
+AZURE_RESOURCE_GROUP ?= gpu-provisioner-amd-rg
+AZURE_CLUSTER_NAME ?= gpu-provisioner-amd-aks
+AZURE_LOCATION ?= eastus2
+IDENTITY_NAME ?= gpu-provisioner-amd-id
+
+.PHONY: all build nodeagent unit-test gpu-test e2etests bench lint clean \
+        docker-build helm-template az-mkrg az-mkaks az-identity az-federated-credential az-patch-helm
+
+all: build
+
+build: nodeagent
+
+nodeagent: gpu_provisioner_amd/_native/libmi355x_nodeagent.so
+
+gpu_provisioner_amd/_native/libmi355x_nodeagent.so: nodeagent/agent.hip
+	mkdir -p gpu_provisioner_amd/_native
+	$(HIPCC) --offload-arch=$(OFFLOAD_ARCH) -O3 -shared -fPIC $< -o $@
+
+unit-test:
+	$(PYTHON) -m pytest tests/ -q -m "not gpu"
+
+gpu-test: nodeagent
+	$(PYTHON) -m pytest tests/ -q -m gpu
+
+# the in-process e2e suite (full controller topology against the AKS simulator)
+e2etests:
+	$(PYTHON) -m pytest tests/test_e2e_suite.py -q
+
+bench:
+	$(PYTHON) bench.py --steps 10 --warmup 3
+
+lint:
+	$(PYTHON) -m py_compile $$(git ls-files '*.py')
+
+docker-build:
+	docker build --build-arg VERSION=$(VERSION) -t $(IMAGE) .
+
+helm-template:
+	helm template gpu-provisioner-amd charts/gpu-provisioner-amd \
+	  --namespace gpu-provisioner \
+	  --set settings.azure.clusterName=$(AZURE_CLUSTER_NAME)
+
+clean:
+	rm -rf gpu_provisioner_amd/_native/*.so build/ .pytest_cache
+
+# --- Azure cluster/identity setup (reference Makefile:62-122) ---------------
+
+az-mkrg:
+	az group create --name $(AZURE_RESOURCE_GROUP) --location $(AZURE_LOCATION)
+
+az-mkaks:
+	az aks create --name $(AZURE_CLUSTER_NAME) --resource-group $(AZURE_RESOURCE_GROUP) \
+	  --node-count 1 --generate-ssh-keys --enable-oidc-issuer --enable-workload-identity
+
+az-identity:
+	az identity create --name $(IDENTITY_NAME) --resource-group $(AZURE_RESOURCE_GROUP)
+	az role assignment create \
+	  --assignee $$(az identity show --name $(IDENTITY_NAME) --resource-group $(AZURE_RESOURCE_GROUP) --query principalId -o tsv) \
+	  --role "Contributor" \
+	  --scope /subscriptions/$(AZURE_SUBSCRIPTION_ID)/resourceGroups/$(AZURE_RESOURCE_GROUP)
+
+az-federated-credential:
+	az identity federated-credential create --name gpu-provisioner-amd \
+	  --identity-name $(IDENTITY_NAME) --resource-group $(AZURE_RESOURCE_GROUP) \
+	  --issuer $$(az aks show --name $(AZURE_CLUSTER_NAME) --resource-group $(AZURE_RESOURCE_GROUP) --query oidcIssuerProfile.issuerUrl -o tsv) \
+	  --subject system:serviceaccount:gpu-provisioner:gpu-provisioner-amd \
+	  --audiences api://AzureADTokenExchange
+
+az-patch-helm:
+	./hack/deploy/configure-helm-values.sh $(AZURE_CLUSTER_NAME) $(AZURE_RESOURCE_GROUP) $(IDENTITY_NAME)

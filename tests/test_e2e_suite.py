@@ -1,0 +1,246 @@
+"""The e2e suite: the reference's 8 ginkgo scenarios
+(test/e2e/suites/suite_test.go) run against the full in-process controller
+topology + AKS simulator. Each spec drives only the public surface
+(NodeClaim/Node objects) and asserts cluster-visible outcomes, like the
+reference's Environment/Eventually harness."""
+import asyncio
+
+import pytest
+
+from gpu_provisioner_amd.apis import v1 as karpv1
+from gpu_provisioner_amd.apis import v1alpha1
+from gpu_provisioner_amd.fake.harness import Harness
+from gpu_provisioner_amd.kube import objects as ko
+from tests.conftest import run
+
+
+def env(**kw) -> Harness:
+    return Harness(
+        ready_latency=0.05, plugin_latency=0.05, **kw
+    ).add_all_controllers(gc_interval=60.0)
+
+
+def spec_nodeclaim(name, labels, vm="Standard_ND128isr_MI355X_v6", node_class=None, annotations=None):
+    nc = karpv1.new_nodeclaim(name, labels=labels)
+    nc["spec"] = {
+        "requirements": [
+            {"key": karpv1.INSTANCE_TYPE_LABEL_KEY, "operator": "In", "values": [vm]}
+        ],
+        "resources": {"requests": {karpv1.AMD_GPU_RESOURCE: "8"}},
+        "nodeClassRef": node_class
+        or {"group": "kaito.sh", "kind": "KaitoNodeClass", "name": "default"},
+    }
+    if annotations:
+        nc["metadata"]["annotations"] = annotations
+    return nc
+
+
+def test_spec1_provision_via_workspace_label():
+    async def main():
+        h = env()
+        await h.start()
+        try:
+            await h.kube.create(spec_nodeclaim("ws1", {karpv1.KAITO_WORKSPACE_LABEL_KEY: "llm"}))
+            nc = await h.wait_initialized("ws1")
+            node = await h.kube.get("v1", "Node", nc["status"]["nodeName"])
+            assert ko.node_is_ready(node)
+            assert ko.node_allocatable(node)[karpv1.AMD_GPU_RESOURCE] == "8"
+            assert ko.labels_of(node)[karpv1.KAITO_WORKSPACE_LABEL_KEY] == "llm"
+        finally:
+            await h.stop()
+
+    run(main())
+
+
+def test_spec2_provision_via_ragengine_label():
+    async def main():
+        h = env()
+        await h.start()
+        try:
+            await h.kube.create(spec_nodeclaim("rag1", {karpv1.KAITO_RAGENGINE_LABEL_KEY: "rag"}))
+            nc = await h.wait_initialized("rag1")
+            assert karpv1.is_initialized(nc)
+        finally:
+            await h.stop()
+
+    run(main())
+
+
+def test_spec3_terminate_via_nodeclaim_delete():
+    async def main():
+        h = env()
+        await h.start()
+        try:
+            await h.kube.create(spec_nodeclaim("del1", {karpv1.KAITO_WORKSPACE_LABEL_KEY: "x"}))
+            nc = await h.wait_initialized("del1")
+            node_name = nc["status"]["nodeName"]
+            await h.kube.delete(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "del1")
+            await h.wait_gone(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "del1")
+            await h.wait_gone("v1", "Node", node_name)
+            assert "del1" not in h.agent_pools.pools
+        finally:
+            await h.stop()
+
+    run(main())
+
+
+def test_spec4_terminate_via_node_delete():
+    async def main():
+        h = env()
+        await h.start()
+        try:
+            await h.kube.create(spec_nodeclaim("del2", {karpv1.KAITO_WORKSPACE_LABEL_KEY: "x"}))
+            nc = await h.wait_initialized("del2")
+            node_name = nc["status"]["nodeName"]
+            await h.kube.delete("v1", "Node", node_name)
+            await h.wait_gone("v1", "Node", node_name)
+            await h.wait_gone(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "del2")
+            assert "del2" not in h.agent_pools.pools
+        finally:
+            await h.stop()
+
+    run(main())
+
+
+def test_spec5_provision_via_kaitonodeclass_ref():
+    """Managed purely by NodeClassRef GroupKind, no kaito labels."""
+
+    async def main():
+        h = env()
+        await h.start()
+        try:
+            await h.kube.create(await_class(h))
+            await h.kube.create(
+                spec_nodeclaim(
+                    "ncref1",
+                    labels={"app": "custom"},
+                    node_class=v1alpha1.node_class_ref("default"),
+                )
+            )
+            nc = await h.wait_initialized("ncref1")
+            assert karpv1.is_initialized(nc)
+        finally:
+            await h.stop()
+
+    run(main())
+
+
+def await_class(h):
+    return v1alpha1.new_kaitonodeclass("default")
+
+
+def test_spec6_negative_foreign_nodeclass_ignored():
+    """AKSNodeClass ref + no kaito labels → no finalizer, no instance, no node
+    (reference suite_test.go:387-450)."""
+
+    async def main():
+        h = env()
+        await h.start()
+        try:
+            await h.kube.create(
+                spec_nodeclaim(
+                    "foreign1",
+                    labels={"app": "other"},
+                    node_class={"group": "karpenter.azure.com", "kind": "AKSNodeClass", "name": "d"},
+                )
+            )
+            await asyncio.sleep(0.5)
+            nc = await h.kube.get(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "foreign1")
+            assert not ko.has_finalizer(nc, karpv1.TERMINATION_FINALIZER)
+            assert not karpv1.is_launched(nc)
+            assert h.agent_pools.create_calls == 0
+            assert len(await h.kube.list("v1", "Node")) == 0
+        finally:
+            await h.stop()
+
+    run(main())
+
+
+def test_spec7_azurelinux_annotation_sets_os_image():
+    async def main():
+        h = env()
+        await h.start()
+        try:
+            await h.kube.create(
+                spec_nodeclaim(
+                    "azlinux1",
+                    labels={karpv1.KAITO_WORKSPACE_LABEL_KEY: "x"},
+                    annotations={karpv1.NODE_IMAGE_FAMILY_ANNOTATION_KEY: "AzureLinux"},
+                )
+            )
+            nc = await h.wait_initialized("azlinux1")
+            assert h.agent_pools.pools["azlinux1"]["properties"]["osSKU"] == "AzureLinux"
+            node = await h.kube.get("v1", "Node", nc["status"]["nodeName"])
+            assert "AzureLinux" in node["status"]["nodeInfo"]["osImage"]
+        finally:
+            await h.stop()
+
+    run(main())
+
+
+def test_spec8_delete_while_provisioning():
+    """Deleting a NodeClaim mid-create must still tear everything down (the
+    GC-covered crash window, reference delete-trigger spec + GC readme)."""
+
+    async def main():
+        h = Harness(create_latency=0.3, ready_latency=0.05).add_all_controllers(
+            gc_interval=0.3, adoption_age=0.1
+        )
+        await h.start()
+        try:
+            await h.kube.create(spec_nodeclaim("mid1", {karpv1.KAITO_WORKSPACE_LABEL_KEY: "x"}))
+
+            async def create_started():
+                return h.agent_pools.create_calls > 0 or None
+
+            await h.wait_for(create_started)
+            # delete while the agent-pool LRO is still running
+            await h.kube.delete(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "mid1")
+            await h.wait_gone(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "mid1", timeout=20)
+
+            async def pool_gone():
+                return "mid1" not in h.agent_pools.pools or None
+
+            await h.wait_for(pool_gone, timeout=20)
+            # no leaked nodes either
+            for node in await h.kube.list("v1", "Node"):
+                assert ko.labels_of(node).get(karpv1.AGENTPOOL_LABEL_KEY) != "mid1"
+        finally:
+            await h.stop()
+
+    run(main())
+
+
+def test_spec9_workload_pod_binds_to_provisioned_node():
+    """BASELINE config #4: a workload pod requesting amd.com/gpu schedules
+    onto the provisioned MI355X node (binding simulated at the apiserver)."""
+
+    async def main():
+        h = env()
+        await h.start()
+        try:
+            await h.kube.create(spec_nodeclaim("bind1", {karpv1.KAITO_WORKSPACE_LABEL_KEY: "x"}))
+            nc = await h.wait_initialized("bind1")
+            node_name = nc["status"]["nodeName"]
+            node = await h.kube.get("v1", "Node", node_name)
+            # the scheduler's feasibility view: allocatable has 8 amd.com/gpu
+            assert ko.qty(ko.node_allocatable(node)[karpv1.AMD_GPU_RESOURCE]) == ko.qty("8")
+            pod = {
+                "apiVersion": "v1",
+                "kind": "Pod",
+                "metadata": {"name": "train", "namespace": "default"},
+                "spec": {
+                    "nodeName": node_name,
+                    "containers": [
+                        {"name": "c", "resources": {"requests": {karpv1.AMD_GPU_RESOURCE: "8"}}}
+                    ],
+                },
+                "status": {"phase": "Running"},
+            }
+            await h.kube.create(pod)
+            got = await h.kube.get("v1", "Pod", "train", "default")
+            assert got["spec"]["nodeName"] == node_name
+        finally:
+            await h.stop()
+
+    run(main())
