@@ -54,14 +54,14 @@ async def one_step(h: Harness, step: int, concurrent: int, latencies: list) -> N
     async def provision(name: str) -> None:
         t0 = time.monotonic()
         await h.kube.create(h.make_nodeclaim(name, VM_SIZE))
-        await h.wait_initialized(name, timeout=60.0)
+        await h.wait_initialized(name, timeout=60.0, interval=0.001)
         latencies.append(time.monotonic() - t0)
 
     await asyncio.gather(*(provision(n) for n in names))
 
     async def teardown(name: str) -> None:
         await h.kube.delete(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, name)
-        await h.wait_gone(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, name, timeout=60.0)
+        await h.wait_gone(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, name, timeout=60.0, interval=0.001)
 
     await asyncio.gather(*(teardown(n) for n in names))
 

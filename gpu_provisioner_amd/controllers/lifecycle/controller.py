@@ -388,9 +388,10 @@ class LifecycleController:
         if not provider_id:
             return []
         if self.nodes.has_synced:
-            found = self.nodes.by_index("providerID", provider_id)
-            if found:
-                return found
+            # trust the index: node arrival triggers a mapped reconcile, and
+            # the registration requeue is the backstop — a full List per
+            # reconcile is O(cluster) and dominated the provision profile
+            return self.nodes.by_index("providerID", provider_id)
         return [
             n
             for n in await self.kube.list("v1", "Node")

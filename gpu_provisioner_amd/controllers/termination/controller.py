@@ -82,6 +82,9 @@ class TerminationController:
             # reference scales 100-5000 with CPU (termination/controller.go:58-61)
             workers=workers if workers is not None else linear_scale_reconciles(32, 512),
         )
+        nodeclaims.add_index(
+            "providerID", lambda o: o.get("status", {}).get("providerID") or None
+        )
         pods.add_index("nodeName", lambda o: o.get("spec", {}).get("nodeName") or None)
         if volumeattachments is not None:
             volumeattachments.add_index(
@@ -264,11 +267,14 @@ class TerminationController:
         pid = ko.provider_id_of(node)
         if not pid:
             return None
-        claims = [
-            nc
-            for nc in await self.kube.list(karpv1.API_VERSION, karpv1.KIND_NODECLAIM)
-            if karpv1.provider_id_of(nc) == pid
-        ]
+        if self.nodeclaims.has_synced and self.nodeclaims.has_index("providerID"):
+            claims = self.nodeclaims.by_index("providerID", pid)
+        else:
+            claims = [
+                nc
+                for nc in await self.kube.list(karpv1.API_VERSION, karpv1.KIND_NODECLAIM)
+                if karpv1.provider_id_of(nc) == pid
+            ]
         # duplicates → no single source of truth (reference controller.go:101-105)
         return claims[0] if len(claims) == 1 else None
 

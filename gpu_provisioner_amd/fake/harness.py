@@ -174,7 +174,9 @@ class Harness:
                 raise TimeoutError("condition not met within timeout")
             await asyncio.sleep(interval)
 
-    async def wait_initialized(self, name: str, timeout: float = 10.0) -> dict:
+    async def wait_initialized(
+        self, name: str, timeout: float = 10.0, interval: float = 0.01
+    ) -> dict:
         async def check():
             try:
                 nc = await self.kube.get(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, name)
@@ -182,9 +184,12 @@ class Harness:
                 return None
             return nc if karpv1.is_initialized(nc) else None
 
-        return await self.wait_for(check, timeout)
+        return await self.wait_for(check, timeout, interval)
 
-    async def wait_gone(self, api_version: str, kind: str, name: str, timeout: float = 10.0):
+    async def wait_gone(
+        self, api_version: str, kind: str, name: str, timeout: float = 10.0,
+        interval: float = 0.01,
+    ):
         async def check():
             try:
                 await self.kube.get(api_version, kind, name)
@@ -192,4 +197,4 @@ class Harness:
             except Exception:
                 return True
 
-        return await self.wait_for(check, timeout)
+        return await self.wait_for(check, timeout, interval)

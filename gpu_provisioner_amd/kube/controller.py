@@ -56,7 +56,13 @@ class Controller:
         self.name = name
         self.reconcile = reconcile
         self.workers = workers
-        self.queue = RateLimitingQueue(rate_limiter, name=name)
+        # Controller default: client-go's 10 qps global bucket throttles
+        # conflict-retry storms into seconds of idle; karpenter raises it and
+        # so do we (per-item exponential backoff still applies).
+        self.queue = RateLimitingQueue(
+            rate_limiter or RateLimiter(base=0.005, cap=30.0, qps=500.0, burst=2000),
+            name=name,
+        )
         self._tasks: list = []
 
     async def enqueue(self, key: str) -> None:

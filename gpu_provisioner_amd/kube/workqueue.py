@@ -4,6 +4,13 @@ Replaces k8s.io/client-go/util/workqueue (used by every controller in the
 reference via controller-runtime): deduplication (an item is queued at most
 once; re-adds during processing re-queue after Done), delayed adds, per-item
 exponential backoff plus a global token bucket, and retry accounting.
+
+Concurrency note: all state mutations are synchronous (no awaits) on one
+event loop, so no lock is needed; waiters park on an Event. The earlier
+Condition-based design deadlocked under load — asyncio.wait_for() around
+Condition.wait() can consume a notify() while cancelling a timed waiter, and
+a single-notify wakeup delivered to that waiter is lost forever. Event.set()
+wakes every waiter, which cannot lose wakeups.
 """
 from __future__ import annotations
 
@@ -81,60 +88,62 @@ class RateLimitingQueue:
         self._processing: set = set()
         self._delayed: list = []  # heap of (ready_at, seq, item)
         self._seq = 0
-        self._cond = asyncio.Condition()
+        self._wakeup = asyncio.Event()
         self._shutdown = False
         self.adds = 0  # metric: total adds
 
     # -- core ---------------------------------------------------------------
 
     async def add(self, item: Hashable) -> None:
-        async with self._cond:
-            if self._shutdown or item in self._dirty:
-                return
-            self.adds += 1
-            self._dirty.add(item)
-            if item not in self._processing:
-                self._queue.append(item)
-                self._cond.notify()
+        if self._shutdown or item in self._dirty:
+            return
+        self.adds += 1
+        self._dirty.add(item)
+        if item not in self._processing:
+            self._queue.append(item)
+            self._wakeup.set()
 
     async def add_after(self, item: Hashable, delay: float) -> None:
         if delay <= 0:
             await self.add(item)
             return
-        async with self._cond:
-            if self._shutdown:
-                return
-            self._seq += 1
-            heapq.heappush(self._delayed, (time.monotonic() + delay, self._seq, item))
-            self._cond.notify()
+        if self._shutdown:
+            return
+        self._seq += 1
+        heapq.heappush(self._delayed, (time.monotonic() + delay, self._seq, item))
+        self._wakeup.set()  # a sleeping getter must recompute its timeout
 
     async def add_rate_limited(self, item: Hashable) -> None:
         await self.add_after(item, self.rate_limiter.when(item))
 
     async def get(self) -> Any:
         """Block until an item is ready; marks it processing. Returns None on shutdown."""
-        async with self._cond:
-            while True:
-                self._drain_delayed()
-                if self._queue:
-                    item = self._queue.pop(0)
-                    self._dirty.discard(item)
-                    self._processing.add(item)
-                    return item
-                if self._shutdown:
-                    return None
-                timeout = self._next_delay()
+        while True:
+            self._drain_delayed()
+            if self._queue:
+                item = self._queue.pop(0)
+                self._dirty.discard(item)
+                self._processing.add(item)
+                return item
+            if self._shutdown:
+                return None
+            timeout = self._next_delay()
+            # clear-then-wait: any add() after the clear sets the event, so a
+            # wakeup between our empty-queue check and the wait is never lost
+            self._wakeup.clear()
+            if timeout is None:
+                await self._wakeup.wait()
+            else:
                 try:
-                    await asyncio.wait_for(self._cond.wait(), timeout)
+                    await asyncio.wait_for(self._wakeup.wait(), timeout)
                 except asyncio.TimeoutError:
                     pass
 
     async def done(self, item: Hashable) -> None:
-        async with self._cond:
-            self._processing.discard(item)
-            if item in self._dirty:
-                self._queue.append(item)
-                self._cond.notify()
+        self._processing.discard(item)
+        if item in self._dirty:
+            self._queue.append(item)
+            self._wakeup.set()
 
     def forget(self, item: Hashable) -> None:
         self.rate_limiter.forget(item)
@@ -143,9 +152,8 @@ class RateLimitingQueue:
         return self.rate_limiter.num_requeues(item)
 
     async def shutdown(self) -> None:
-        async with self._cond:
-            self._shutdown = True
-            self._cond.notify_all()
+        self._shutdown = True
+        self._wakeup.set()
 
     # -- helpers ------------------------------------------------------------
 
