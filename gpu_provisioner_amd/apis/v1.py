@@ -77,6 +77,7 @@ COND_INITIALIZED = "Initialized"
 COND_DRAINED = "Drained"
 COND_VOLUMES_DETACHED = "VolumesDetached"
 COND_INSTANCE_TERMINATING = "InstanceTerminating"
+COND_DRIFTED = "Drifted"
 COND_READY = "Ready"
 
 # -- NodeClaim accessors ------------------------------------------------------

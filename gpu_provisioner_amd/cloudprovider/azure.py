@@ -15,11 +15,17 @@ from typing import Optional
 from ..apis import v1 as karpv1
 from ..apis import v1alpha1
 from ..kube import objects as ko
+from ..providers.instance import bootstrap
 from ..providers.instance.provider import InstanceProvider
 from ..providers.instancetype.catalog import InstanceTypeProvider
-from .types import CloudProvider, Instance, RepairPolicy
+from .types import CloudProvider, Instance, NodeClaimNotFoundError, RepairPolicy
 
 NODE_REPAIR_TOLERATION_SECONDS = 600.0  # 10 min
+
+# drift reasons (net-new: the reference's IsDrifted is a stub returning "")
+DRIFT_INSTANCE_TYPE = "InstanceTypeDrift"
+DRIFT_NODE_IMAGE = "NodeImageDrift"
+DRIFT_SKU_RETIRED = "SKURetiredDrift"
 
 
 class AzureCloudProvider(CloudProvider):
@@ -50,7 +56,39 @@ class AzureCloudProvider(CloudProvider):
     async def get_instance_types(self, nodepool: Optional[dict] = None) -> list:
         return self.catalog.list()
 
-    def is_drifted(self, nodeclaim: dict) -> str:
+    async def is_drifted(self, nodeclaim: dict) -> str:
+        """Compare the live agent pool against the NodeClaim's declared shape.
+
+        Net improvement over the reference (stub, cloudprovider.go:94-97):
+          * InstanceTypeDrift — the pool's VM size is no longer among the
+            NodeClaim's node.kubernetes.io/instance-type requirement values;
+          * NodeImageDrift — the pool's osSKU disagrees with what the
+            kaito.sh/node-image-family annotation selects today;
+          * SKURetiredDrift — the pool runs an AMD GPU SKU that has been
+            retired from the MI355X catalog (no longer orderable).
+        A vanished instance is NOT drift — that is the GC controllers' domain.
+        """
+        pid = karpv1.provider_id_of(nodeclaim)
+        if not pid:
+            return ""
+        try:
+            instance = await self.instances.get(pid)
+        except NodeClaimNotFoundError:
+            return ""
+        wanted = karpv1.requirement_values(nodeclaim, karpv1.INSTANCE_TYPE_LABEL_KEY)
+        if wanted and instance.type and instance.type not in wanted:
+            return DRIFT_INSTANCE_TYPE
+        want_sku = bootstrap.determine_os_sku(
+            ko.annotations_of(nodeclaim).get(karpv1.NODE_IMAGE_FAMILY_ANNOTATION_KEY, "")
+        )
+        if instance.os_sku and instance.os_sku != want_sku:
+            return DRIFT_NODE_IMAGE
+        if (
+            instance.type
+            and self.catalog.get(instance.type) is None
+            and self.catalog.is_gpu_sku(instance.type)
+        ):
+            return DRIFT_SKU_RETIRED
         return ""
 
     def repair_policies(self) -> list:

@@ -64,8 +64,8 @@ class MetricsDecorator(CloudProvider):
     async def get_instance_types(self, nodepool: Optional[dict] = None) -> list:
         return await self._call("GetInstanceTypes", self.inner.get_instance_types(nodepool))
 
-    def is_drifted(self, nodeclaim: dict) -> str:
-        return self.inner.is_drifted(nodeclaim)
+    async def is_drifted(self, nodeclaim: dict) -> str:
+        return await self._call("IsDrifted", self.inner.is_drifted(nodeclaim))
 
     def repair_policies(self) -> list:
         return self.inner.repair_policies()

@@ -133,6 +133,7 @@ class Instance:
     labels: dict = field(default_factory=dict)
     tags: dict = field(default_factory=dict)
     created_at: str = ""
+    os_sku: str = ""  # pool osSKU (Ubuntu/AzureLinux) — drift detection input
 
 
 @dataclass
@@ -174,8 +175,10 @@ class CloudProvider(abc.ABC):
         """Orderable InstanceTypes (the MI355X catalog)."""
 
     @abc.abstractmethod
-    def is_drifted(self, nodeclaim: dict) -> str:
-        """Drift reason or '' (reference returns always-empty, cloudprovider.go:94-97)."""
+    async def is_drifted(self, nodeclaim: dict) -> str:
+        """Drift reason or ''. The reference stubs this to always-empty
+        (cloudprovider.go:94-97); here it compares the live agent pool against
+        the NodeClaim's declared shape (see azure.py)."""
 
     @abc.abstractmethod
     def repair_policies(self) -> list:

@@ -82,9 +82,13 @@ class Harness:
         gc_interval: float = 0.5,
         adoption_age: float = 0.2,
         with_health: bool = True,
+        with_drift: bool = True,
+        drift_interval: Optional[float] = None,
+        drift_replace: bool = False,
     ) -> "Harness":
         """Wire the full controller set (the production main() topology) with
         test-friendly cadences."""
+        from ..controllers.drift.controller import DriftController
         from ..controllers.garbagecollection.controller import (
             InstanceGCController,
             NodeClaimGCController,
@@ -120,6 +124,13 @@ class Harness:
                 self.kube, self.cloud, self.recorder, self.nodes
             )
             self.controllers.append(self.health)
+        if with_drift:
+            self.drift = DriftController(
+                self.kube, self.cloud, self.recorder,
+                interval=drift_interval if drift_interval is not None else gc_interval,
+                replace=drift_replace,
+            )
+            self.controllers.append(self.drift)
         return self
 
     # -- lifecycle -----------------------------------------------------------

@@ -15,6 +15,7 @@ import logging
 
 from .apis import v1 as karpv1
 from .cloudprovider.decorator import MetricsDecorator
+from .controllers.drift.controller import DriftController
 from .controllers.garbagecollection.controller import (
     InstanceGCController,
     NodeClaimGCController,
@@ -59,6 +60,12 @@ def build_manager(kube, options: Options, cloud_provider, version: str = "0.1.0"
     # (reference vendor/.../controllers/controllers.go:109-111)
     if options.feature_gates.node_repair and cloud.repair_policies():
         controllers.append(HealthController(kube, cloud, recorder, nodes))
+    # drift detection (net-new: the reference stubs IsDrifted); replacement
+    # only when DriftReplace is explicitly gated on
+    if options.feature_gates.drift:
+        controllers.append(
+            DriftController(kube, cloud, recorder, replace=options.feature_gates.drift_replace)
+        )
 
     manager = Manager(
         kube,

@@ -73,6 +73,12 @@ NODECLAIMS_DISRUPTED = _get_or_create(
     "Number of nodeclaims disrupted",
     ("reason", "nodepool"),
 )
+NODECLAIMS_DRIFTED = _get_or_create(
+    Counter,
+    "karpenter_nodeclaims_drifted_total",
+    "Number of nodeclaims detected as drifted from their spec",
+    ("reason", "nodepool"),
+)
 NODES_CREATED = _get_or_create(
     Counter,
     "karpenter_nodes_created_total",

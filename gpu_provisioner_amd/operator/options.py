@@ -16,6 +16,13 @@ from dataclasses import dataclass, field
 @dataclass
 class FeatureGates:
     node_repair: bool = True
+    # Drift marks NodeClaims whose agent pool no longer matches their spec
+    # (Drifted condition); DriftReplace additionally deletes them so the
+    # owner (KAITO) re-creates a conforming node. Detection defaults on,
+    # replacement defaults off (parity with the reference, which has no
+    # active disruption controllers).
+    drift: bool = True
+    drift_replace: bool = False
 
     @classmethod
     def parse(cls, s: str) -> "FeatureGates":
@@ -26,8 +33,13 @@ class FeatureGates:
                 continue
             name, val = part.split("=", 1)
             enabled = val.strip().lower() == "true"
-            if name.strip() == "NodeRepair":
+            name = name.strip()
+            if name == "NodeRepair":
                 gates.node_repair = enabled
+            elif name == "Drift":
+                gates.drift = enabled
+            elif name == "DriftReplace":
+                gates.drift_replace = enabled
         return gates
 
 

@@ -282,4 +282,5 @@ class InstanceProvider:
             labels=dict(labels),
             tags=dict(props.get("tags") or {}),
             created_at=labels.get(CREATION_TIMESTAMP_LABEL, ""),
+            os_sku=props.get("osSKU", ""),
         )

@@ -211,6 +211,65 @@ def test_spec8_delete_while_provisioning():
     run(main())
 
 
+def test_spec10_32_concurrent_churn_with_drift():
+    """BASELINE config #5: 32 concurrent NodeClaims with drift detection +
+    delete/disruption reconcile under churn. One third of the fleet is
+    mutated out-of-band (drift → detected), one third deleted mid-life
+    (churn), the rest must stay Initialized and un-drifted throughout."""
+
+    async def main():
+        h = Harness(node_wait_interval=0.002).add_all_controllers(
+            lifecycle_workers=64, gc_interval=60.0, drift_interval=0.05,
+            drift_replace=True,
+        )
+        await h.start()
+        try:
+            names = [f"churn{i:02d}" for i in range(32)]
+            await asyncio.gather(
+                *(
+                    h.kube.create(
+                        spec_nodeclaim(n, {karpv1.KAITO_WORKSPACE_LABEL_KEY: "fleet"})
+                    )
+                    for n in names
+                )
+            )
+            await asyncio.gather(*(h.wait_initialized(n, timeout=30) for n in names))
+
+            drifting, deleting, steady = names[:10], names[10:20], names[20:]
+            for n in drifting:
+                h.agent_pools.pools[n]["properties"]["osSKU"] = "AzureLinux"
+            await asyncio.gather(
+                *(h.kube.delete(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, n) for n in deleting)
+            )
+            # drifted claims are detected and (DriftReplace on) replaced:
+            # the claim is deleted and its pool torn down, concurrently with
+            # the explicit churn deletes
+            await asyncio.gather(
+                *(
+                    h.wait_gone(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, n, timeout=30)
+                    for n in drifting + deleting
+                )
+            )
+
+            async def pools_gone():
+                return (
+                    all(n not in h.agent_pools.pools for n in drifting + deleting) or None
+                )
+
+            await h.wait_for(pools_gone, timeout=30)
+            # the steady fleet survived the churn untouched
+            for n in steady:
+                nc = await h.kube.get(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, n)
+                assert karpv1.is_initialized(nc)
+                assert not ko.is_deleting(nc)
+                assert not ko.condition_is_true(nc, karpv1.COND_DRIFTED)
+                assert n in h.agent_pools.pools
+        finally:
+            await h.stop()
+
+    run(main())
+
+
 def test_spec9_workload_pod_binds_to_provisioned_node():
     """BASELINE config #4: a workload pod requesting amd.com/gpu schedules
     onto the provisioned MI355X node (binding simulated at the apiserver)."""

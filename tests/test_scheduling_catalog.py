@@ -157,7 +157,7 @@ def test_metrics_decorator_counts_errors_and_duration():
         async def get(self, pid): return {}
         async def list(self): return []
         async def get_instance_types(self, np=None): return []
-        def is_drifted(self, nc): return ""
+        async def is_drifted(self, nc): return ""
         def repair_policies(self): return []
         def name(self): return "test-provider"
         def get_supported_node_classes(self): return []
