@@ -55,6 +55,10 @@ class GPUReport:
     hbm_bw_gbs: float = 0.0
     fma_ok: bool = False
     mfma_ok: bool = False
+    mfma_bf16_ok: bool = False
+    mfma_fp8_ok: bool = False
+    lds_ok: bool = False
+    lds_bytes_tested: int = 0
     healthy: bool = False
     problems: list = field(default_factory=list)
 
@@ -111,6 +115,20 @@ class NodeAgent:
     def mfma_selftest(self, dev: int) -> bool:
         return self.lib.na_mfma_selftest(dev) == 0
 
+    def mfma_bf16_selftest(self, dev: int) -> bool:
+        """v_mfma_f32_16x16x32_bf16 — the CDNA4 training datapath."""
+        return self.lib.na_mfma_bf16_selftest(dev) == 0
+
+    def mfma_fp8_selftest(self, dev: int) -> bool:
+        """v_mfma_f32_16x16x32_fp8_fp8 (E4M3) — the serving datapath."""
+        return self.lib.na_mfma_fp8_selftest(dev) == 0
+
+    def lds_selftest(self, dev: int) -> tuple:
+        """(ok, bytes_tested): whole-LDS pattern write/swizzled-read check."""
+        tested = ctypes.c_longlong(0)
+        rc = self.lib.na_lds_selftest(dev, ctypes.byref(tested))
+        return rc == 0, tested.value
+
     def p2p_matrix(self, n: int) -> list:
         buf = (ctypes.c_int * (n * n))()
         if self.lib.na_p2p_matrix(n, buf) != 0:
@@ -153,6 +171,15 @@ class NodeAgent:
                 g.mfma_ok = self.mfma_selftest(d)
                 if not g.mfma_ok:
                     g.problems.append(f"MFMA matrix-pipe selftest failed: {self._err()}")
+                g.mfma_bf16_ok = self.mfma_bf16_selftest(d)
+                if not g.mfma_bf16_ok:
+                    g.problems.append(f"MFMA bf16 selftest failed: {self._err()}")
+                g.mfma_fp8_ok = self.mfma_fp8_selftest(d)
+                if not g.mfma_fp8_ok:
+                    g.problems.append(f"MFMA fp8 selftest failed: {self._err()}")
+                g.lds_ok, g.lds_bytes_tested = self.lds_selftest(d)
+                if not g.lds_ok:
+                    g.problems.append(f"LDS selftest failed: {self._err()}")
             except NodeAgentError as e:
                 g.problems.append(str(e))
             g.healthy = not g.problems

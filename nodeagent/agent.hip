@@ -79,6 +79,64 @@ __global__ void mfma_selftest_kernel(float* __restrict__ out, float alpha, float
 #endif
 }
 
+// MFMA datatype-path self-tests for the pipes ML workloads actually run on:
+// gfx950's 2xK bf16 form (v_mfma_f32_16x16x32_bf16, the CDNA4 training
+// workhorse) and the fp8 E4M3 form (v_mfma_f32_16x16x32_fp8_fp8, the serving
+// path). Same uniform-operand trick as the f32 test: with A=alpha and B=beta
+// everywhere and C=0, every output element is K*alpha*beta regardless of
+// fragment layout, so the check is layout-independent while still exercising
+// the low-precision input muxes and the accumulator file.
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+
+__global__ void mfma_bf16_selftest_kernel(float* __restrict__ out, float alpha, float beta) {
+#if defined(__gfx950__)
+    bf16x8 a, b;
+    for (int i = 0; i < 8; ++i) {
+        a[i] = (__bf16)alpha;
+        b[i] = (__bf16)beta;
+    }
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+    int lane = threadIdx.x;
+    for (int i = 0; i < 4; ++i) out[lane * 4 + i] = acc[i];
+#else
+    out[threadIdx.x] = -1.0f;
+#endif
+}
+
+// fp8 operands are 8 packed E4M3 bytes per lane (one i64).
+__global__ void mfma_fp8_selftest_kernel(float* __restrict__ out, long a_bits, long b_bits) {
+#if defined(__gfx950__)
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a_bits, b_bits, acc, 0, 0, 0);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a_bits, b_bits, acc, 0, 0, 0);
+    int lane = threadIdx.x;
+    for (int i = 0; i < 4; ++i) out[lane * 4 + i] = acc[i];
+#else
+    out[threadIdx.x] = -1.0f;
+#endif
+}
+
+// LDS self-test: fill the whole per-WG allocation with a position-dependent
+// pattern, barrier, read back through a bank-swizzled index. Exercises the
+// LDS array + crossbar across all CUs (one WG per CU's worth of a big grid).
+__global__ void lds_selftest_kernel(unsigned* __restrict__ fails, size_t words) {
+    extern __shared__ unsigned lds[];
+    unsigned t = threadIdx.x, n = blockDim.x;
+    for (size_t i = t; i < words; i += n)
+        lds[i] = (unsigned)(i * 2654435761u) ^ (blockIdx.x * 97u);
+    __syncthreads();
+    unsigned bad = 0;
+    for (size_t i = t; i < words; i += n) {
+        // swizzle: XOR within a 64-dword bank row so every lane group hits
+        // addresses written by other lanes (crossbar coverage)
+        size_t j = (i & ~(size_t)63) | ((i ^ 37) & 63);
+        if (j < words && lds[j] != ((unsigned)(j * 2654435761u) ^ (blockIdx.x * 97u))) bad++;
+    }
+    if (bad) atomicAdd(fails, bad);
+}
+
 // ---------------------------------------------------------------------------
 // C ABI
 // ---------------------------------------------------------------------------
@@ -178,6 +236,85 @@ extern "C" int na_mfma_selftest(int dev) {
                           "mfma selftest: elem %d got %g want %g", i, host[i], want);
             return NA_ERR_VERIFY;
         }
+    }
+    return NA_OK;
+}
+
+// Shared verify for the 16x16x32 family: two chained MFMAs, K=32, uniform
+// operands → every element = 2*K*a*b.
+static int verify_mfma_out(const char* which, const float* host, int n, float want) {
+    for (int i = 0; i < n; ++i) {
+        if (host[i] != want) {
+            std::snprintf(na_last_error_buf, sizeof(na_last_error_buf),
+                          "%s selftest: elem %d got %g want %g", which, i, host[i], want);
+            return NA_ERR_VERIFY;
+        }
+    }
+    return NA_OK;
+}
+
+extern "C" int na_mfma_bf16_selftest(int dev) {
+    HIP_CHECK(hipSetDevice(dev));
+    // 1.5 and 2.0 are exact in bf16; 2*32*1.5*2.0 = 192 exact in f32
+    const float alpha = 1.5f, beta = 2.0f;
+    const int n = 64 * 4;
+    float* out = nullptr;
+    HIP_CHECK(hipMalloc(&out, n * sizeof(float)));
+    mfma_bf16_selftest_kernel<<<dim3(1), dim3(64)>>>(out, alpha, beta);
+    HIP_CHECK(hipDeviceSynchronize());
+    float host[n];
+    HIP_CHECK(hipMemcpy(host, out, n * sizeof(float), hipMemcpyDeviceToHost));
+    (void)hipFree(out);
+    return verify_mfma_out("mfma-bf16", host, n, 2.0f * 32.0f * alpha * beta);
+}
+
+extern "C" int na_mfma_fp8_selftest(int dev) {
+    HIP_CHECK(hipSetDevice(dev));
+    // E4M3: 1.5 = 0x3C, 2.0 = 0x40 (both exact); 2*32*1.5*2.0 = 192
+    const long a_bits = 0x3C3C3C3C3C3C3C3CLL;
+    const long b_bits = 0x4040404040404040LL;
+    const int n = 64 * 4;
+    float* out = nullptr;
+    HIP_CHECK(hipMalloc(&out, n * sizeof(float)));
+    mfma_fp8_selftest_kernel<<<dim3(1), dim3(64)>>>(out, a_bits, b_bits);
+    HIP_CHECK(hipDeviceSynchronize());
+    float host[n];
+    HIP_CHECK(hipMemcpy(host, out, n * sizeof(float), hipMemcpyDeviceToHost));
+    (void)hipFree(out);
+    return verify_mfma_out("mfma-fp8", host, n, 192.0f);
+}
+
+extern "C" int na_lds_selftest(int dev, long long* bytes_tested) {
+    HIP_CHECK(hipSetDevice(dev));
+    hipDeviceProp_t prop;
+    HIP_CHECK(hipGetDeviceProperties(&prop, dev));
+    // whole per-WG LDS allocation (160 KiB/CU on gfx950, minus any
+    // runtime-reserved slice reported via sharedMemPerBlock)
+    size_t lds_bytes = prop.sharedMemPerBlock;
+    if (lds_bytes > 160 * 1024) lds_bytes = 160 * 1024;
+    size_t words = lds_bytes / sizeof(unsigned);
+    unsigned* fails = nullptr;
+    HIP_CHECK(hipMalloc(&fails, sizeof(unsigned)));
+    HIP_CHECK(hipMemset(fails, 0, sizeof(unsigned)));
+    // 2048 WGs: every CU's LDS array gets exercised multiple times
+    hipError_t e = hipSuccess;
+    lds_selftest_kernel<<<dim3(2048), dim3(256), lds_bytes>>>(fails, words);
+    e = hipGetLastError();
+    if (e != hipSuccess) {
+        (void)hipFree(fails);
+        std::snprintf(na_last_error_buf, sizeof(na_last_error_buf), "lds launch: %s",
+                      hipGetErrorString(e));
+        return NA_ERR_HIP;
+    }
+    HIP_CHECK(hipDeviceSynchronize());
+    unsigned bad = 0;
+    HIP_CHECK(hipMemcpy(&bad, fails, sizeof(unsigned), hipMemcpyDeviceToHost));
+    (void)hipFree(fails);
+    if (bytes_tested) *bytes_tested = (long long)lds_bytes;
+    if (bad) {
+        std::snprintf(na_last_error_buf, sizeof(na_last_error_buf),
+                      "lds selftest: %u mismatched words across %zu-byte LDS", bad, lds_bytes);
+        return NA_ERR_VERIFY;
     }
     return NA_OK;
 }

@@ -32,6 +32,9 @@ def test_hip_library_builds_and_loads():
         "na_hbm_bandwidth",
         "na_fma_selftest",
         "na_mfma_selftest",
+        "na_mfma_bf16_selftest",
+        "na_mfma_fp8_selftest",
+        "na_lds_selftest",
         "na_p2p_matrix",
         "na_p2p_bandwidth",
         "na_last_error",
@@ -84,6 +87,29 @@ def test_gpu_fma_and_mfma_selftests():
     agent = NodeAgent()
     assert agent.fma_selftest(0), "VALU FMA selftest failed"
     assert agent.mfma_selftest(0), "MFMA (v_mfma_f32_16x16x4_f32) selftest failed"
+
+
+@pytest.mark.gpu
+def test_gpu_mfma_datatype_paths():
+    """The CDNA4 low-precision pipes ML workloads run on: bf16 (training)
+    and fp8 E4M3 (serving) 16x16x32 MFMA forms."""
+    from gpu_provisioner_amd.nodeagent import NodeAgent
+
+    _ensure_lib()
+    agent = NodeAgent()
+    assert agent.mfma_bf16_selftest(0), "MFMA bf16 (v_mfma_f32_16x16x32_bf16) failed"
+    assert agent.mfma_fp8_selftest(0), "MFMA fp8 (v_mfma_f32_16x16x32_fp8_fp8) failed"
+
+
+@pytest.mark.gpu
+def test_gpu_lds_selftest():
+    from gpu_provisioner_amd.nodeagent import NodeAgent
+
+    _ensure_lib()
+    agent = NodeAgent()
+    ok, tested = agent.lds_selftest(0)
+    assert ok, "LDS selftest failed"
+    assert tested >= 64 * 1024, f"only {tested} LDS bytes tested"
 
 
 @pytest.mark.gpu
