@@ -81,13 +81,18 @@ class InMemoryAPIServer:
                 raise err
 
     def _broadcast(self, gvk: tuple, event_type: str, obj: dict) -> None:
+        # `obj` is an immutable revision object (see _apply_update): the
+        # store, the history and every watcher share it without copying.
+        # Watch consumers (informers) treat event objects as read-only —
+        # the client-go cache contract — and the server never mutates a
+        # stored revision in place (every write replaces it).
         rv = int(obj["metadata"]["resourceVersion"])
-        self._history.append((rv, gvk, event_type, ko.deep_copy(obj)))
+        self._history.append((rv, gvk, event_type, obj))
         if len(self._history) > _WATCH_HISTORY:
             self._history = self._history[-_WATCH_HISTORY:]
         for wgvk, queue in self._watchers:
             if wgvk == gvk:
-                queue.put_nowait((event_type, ko.deep_copy(obj)))
+                queue.put_nowait((event_type, obj))
 
     def _bump(self, obj: dict) -> None:
         obj["metadata"]["resourceVersion"] = self._next_rv()
@@ -170,8 +175,13 @@ class InMemoryAPIServer:
             return self._apply_update(gvk, key, cur, obj, subresource)
 
     def _apply_update(self, gvk: tuple, key: tuple, cur: dict, obj: dict, subresource: str) -> dict:
-        stored = ko.deep_copy(cur)
+        # Revision objects are immutable: a new revision may SHARE unchanged
+        # subtrees with the previous one (they are never mutated in place),
+        # so a status write is a shallow re-root + one status copy, not a
+        # full-object copy. Caller-provided data is still deep-copied — the
+        # caller retains its reference and may mutate it afterwards.
         if subresource == "status":
+            stored = {**cur, "metadata": {**cur["metadata"]}}
             stored["status"] = ko.deep_copy(obj.get("status", {}))
         else:
             new = ko.deep_copy(obj)
@@ -215,9 +225,8 @@ class InMemoryAPIServer:
             if cur is None:
                 raise NotFoundError(f"{kind} {namespace}/{name} not found")
             if subresource == "status":
-                merged = ko.deep_copy(cur)
-                merged["status"] = json_merge_patch(cur.get("status", {}), patch.get("status", patch))
-                return self._apply_update(gvk, key, cur, merged, "status")
+                merged_status = json_merge_patch(cur.get("status", {}), patch.get("status", patch))
+                return self._apply_update(gvk, key, cur, {"status": merged_status}, "status")
             merged = json_merge_patch(cur, patch)
             # patches may not carry resourceVersion conflicts — merge wins
             merged.setdefault("metadata", {})["resourceVersion"] = cur["metadata"]["resourceVersion"]
@@ -243,13 +252,13 @@ class InMemoryAPIServer:
                 raise ConflictError(f"{kind} {name}: uid precondition failed")
             if ko.finalizers_of(cur):
                 if not ko.is_deleting(cur):
-                    stored = ko.deep_copy(cur)
+                    stored = {**cur, "metadata": {**cur["metadata"]}}
                     ko.meta(stored)["deletionTimestamp"] = ko.fmt_time(ko.now())
                     self._bump(stored)
                     self._store[gvk][key] = stored
                     self._broadcast(gvk, MODIFIED, stored)
                 return
-            stored = ko.deep_copy(cur)
+            stored = {**cur, "metadata": {**cur["metadata"]}}
             ko.meta(stored)["deletionTimestamp"] = ko.fmt_time(ko.now())
             self._bump(stored)
             del self._store[gvk][key]

@@ -185,23 +185,42 @@ class Harness:
                 raise TimeoutError("condition not met within timeout")
             await asyncio.sleep(interval)
 
+    def _informer_for(self, api_version: str, kind: str):
+        for inf in (self.nodeclaims, self.nodes, self.pods, self.volumeattachments):
+            if inf.api_version == api_version and inf.kind == kind:
+                return inf if inf.has_synced else None
+        return None
+
     async def wait_initialized(
         self, name: str, timeout: float = 10.0, interval: float = 0.01
     ) -> dict:
-        async def check():
-            try:
-                nc = await self.kube.get(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, name)
-            except Exception:
-                return None
-            return nc if karpv1.is_initialized(nc) else None
+        """Wait through the watch surface (informer cache — the same view a
+        real client like KAITO observes), falling back to apiserver reads
+        before sync."""
 
-        return await self.wait_for(check, timeout, interval)
+        async def check():
+            inf = self._informer_for(karpv1.API_VERSION, karpv1.KIND_NODECLAIM)
+            if inf is not None:
+                nc = inf.get(name)
+            else:
+                try:
+                    nc = await self.kube.get(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, name)
+                except Exception:
+                    return None
+            return nc if nc and karpv1.is_initialized(nc) else None
+
+        found = await self.wait_for(check, timeout, interval)
+        # private copy: callers may mutate the result (cache objects are shared)
+        return ko.deep_copy(found)
 
     async def wait_gone(
         self, api_version: str, kind: str, name: str, timeout: float = 10.0,
         interval: float = 0.01,
     ):
         async def check():
+            inf = self._informer_for(api_version, kind)
+            if inf is not None:
+                return True if inf.get(name) is None else None
             try:
                 await self.kube.get(api_version, kind, name)
                 return None

@@ -65,18 +65,23 @@ class Informer:
         self._handlers.append(fn)
 
     # -- cache access -------------------------------------------------------
+    #
+    # Shared-object contract (client-go parity): objects returned from the
+    # cache are SHARED and read-only. A caller that wants to mutate must
+    # ko.deep_copy() first — in practice controllers re-read via the kube
+    # client (which returns private copies) before any write. Copying here
+    # was 40% of the provisioning-path profile.
 
     def get(self, name: str, namespace: str = "") -> Optional[dict]:
         key = f"{namespace}/{name}" if namespace else name
-        obj = self._cache.get(key)
-        return ko.deep_copy(obj) if obj else None
+        return self._cache.get(key)
 
     def list(self) -> list:
-        return [ko.deep_copy(o) for o in self._cache.values()]
+        return list(self._cache.values())
 
     def by_index(self, index: str, value: str) -> list:
         _, idx = self._indexes[index]
-        return [ko.deep_copy(self._cache[k]) for k in idx.get(value, set()) if k in self._cache]
+        return [self._cache[k] for k in idx.get(value, set()) if k in self._cache]
 
     async def wait_for_sync(self) -> None:
         await self._synced.wait()

