@@ -53,6 +53,9 @@ EXCLUDE_FROM_LB_LABEL_KEY = "node.kubernetes.io/exclude-from-external-load-balan
 
 # AMD GPU surface (net-new vs the reference's nvidia.com/gpu)
 AMD_GPU_RESOURCE = "amd.com/gpu"
+# Node condition published by the mi355x-nodeagent DaemonSet (HBM/MFMA/LDS/
+# xGMI self-tests); the repair policies watch it like NodeReady
+AMD_GPU_HEALTHY_CONDITION_TYPE = "AMDGPUHealthy"
 AMD_GPU_PRODUCT_LABEL_KEY = "amd.com/gpu.product"
 AMD_GPU_VRAM_LABEL_KEY = "amd.com/gpu.vram"
 AMD_GPU_COUNT_LABEL_KEY = "amd.com/gpu.count"

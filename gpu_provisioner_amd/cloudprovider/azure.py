@@ -21,6 +21,10 @@ from ..providers.instancetype.catalog import InstanceTypeProvider
 from .types import CloudProvider, Instance, NodeClaimNotFoundError, RepairPolicy
 
 NODE_REPAIR_TOLERATION_SECONDS = 600.0  # 10 min
+# GPU-sick nodes (nodeagent's AMDGPUHealthy=False: failed HBM/MFMA/LDS/xGMI
+# self-tests) are repaired faster than NodeReady flaps: the kubelet is fine,
+# the evidence is direct, and a degraded 8x MI355X host is expensive to keep
+GPU_REPAIR_TOLERATION_SECONDS = 300.0  # 5 min
 
 # drift reasons (net-new: the reference's IsDrifted is a stub returning "")
 DRIFT_INSTANCE_TYPE = "InstanceTypeDrift"
@@ -29,12 +33,25 @@ DRIFT_SKU_RETIRED = "SKURetiredDrift"
 
 
 class AzureCloudProvider(CloudProvider):
-    def __init__(self, instances: InstanceProvider, catalog: InstanceTypeProvider):
+    def __init__(
+        self,
+        instances: InstanceProvider,
+        catalog: InstanceTypeProvider,
+        *,
+        repair_toleration: float = NODE_REPAIR_TOLERATION_SECONDS,
+        gpu_repair_toleration: float = GPU_REPAIR_TOLERATION_SECONDS,
+    ):
         self.instances = instances
         self.catalog = catalog
         self._repair_policies = [
-            RepairPolicy("Ready", ko.CONDITION_FALSE, NODE_REPAIR_TOLERATION_SECONDS),
-            RepairPolicy("Ready", ko.CONDITION_UNKNOWN, NODE_REPAIR_TOLERATION_SECONDS),
+            RepairPolicy("Ready", ko.CONDITION_FALSE, repair_toleration),
+            RepairPolicy("Ready", ko.CONDITION_UNKNOWN, repair_toleration),
+            # net-new: on-node GPU evidence from the mi355x-nodeagent
+            RepairPolicy(
+                karpv1.AMD_GPU_HEALTHY_CONDITION_TYPE,
+                ko.CONDITION_FALSE,
+                gpu_repair_toleration,
+            ),
         ]
 
     async def create(self, nodeclaim: dict) -> dict:

@@ -35,6 +35,8 @@ class Harness:
         plugin_latency: float = 0.0,
         node_wait_interval: float = 0.02,
         region: str = "eastus2",
+        repair_toleration: Optional[float] = None,
+        gpu_repair_toleration: Optional[float] = None,
     ):
         self.server = InMemoryAPIServer()
         self.kube = InMemoryClient(self.server)
@@ -59,7 +61,14 @@ class Harness:
             cluster_name="cluster",
             node_wait_interval=node_wait_interval,
         )
-        self.cloud = MetricsDecorator(AzureCloudProvider(self.instances, self.catalog))
+        cloud_kwargs = {}
+        if repair_toleration is not None:
+            cloud_kwargs["repair_toleration"] = repair_toleration
+        if gpu_repair_toleration is not None:
+            cloud_kwargs["gpu_repair_toleration"] = gpu_repair_toleration
+        self.cloud = MetricsDecorator(
+            AzureCloudProvider(self.instances, self.catalog, **cloud_kwargs)
+        )
         self.recorder = EventRecorder(self.kube)
         self.informers = InformerFactory(self.kube)
         self.nodeclaims: Informer = self.informers.informer(karpv1.API_VERSION, karpv1.KIND_NODECLAIM)

@@ -197,11 +197,60 @@ class NodeAgent:
         return report
 
 
+# -- node condition publishing (the repair feedback loop) --------------------
+#
+# The DaemonSet runs the check in a loop and publishes the result as the
+# AMDGPUHealthy condition on its Node. The cloud provider's RepairPolicies
+# include (AMDGPUHealthy, False, 5 min), so the node.health controller
+# replaces nodes with failed HBM/MFMA/LDS/xGMI self-tests — on-node GPU
+# evidence the reference's NodeReady-only repair can't see.
+
+
+def node_condition_from_report(report: NodeReport) -> dict:
+    from .apis import v1 as karpv1
+
+    if report.healthy:
+        return {
+            "type": karpv1.AMD_GPU_HEALTHY_CONDITION_TYPE,
+            "status": "True",
+            "reason": "AllChecksPassed",
+            "message": f"{report.gpu_count} GPU(s) passed HBM/FMA/MFMA/LDS/xGMI self-tests",
+        }
+    return {
+        "type": karpv1.AMD_GPU_HEALTHY_CONDITION_TYPE,
+        "status": "False",
+        "reason": "GPUUnhealthy",
+        "message": "; ".join(report.problems)[:1024] or "node agent found no GPUs",
+    }
+
+
+async def patch_node_condition(kube, node_name: str, report: NodeReport) -> None:
+    """Merge the AMDGPUHealthy condition into the Node's status conditions."""
+    from .kube import objects as ko
+
+    cond = node_condition_from_report(report)
+    node = await kube.get("v1", "Node", node_name)
+    ko.set_condition(node, cond["type"], cond["status"], cond["reason"], cond["message"])
+    await kube.patch(
+        "v1",
+        "Node",
+        node_name,
+        {"status": {"conditions": node["status"]["conditions"]}},
+        subresource="status",
+    )
+
+
 def main(argv=None) -> int:
     ap = argparse.ArgumentParser(description="MI355X node health agent")
     ap.add_argument("--json", action="store_true", help="emit JSON report")
     ap.add_argument("--expect-gpus", type=int, default=0)
     ap.add_argument("--bw-bytes", type=int, default=1 << 30)
+    ap.add_argument(
+        "--patch-node",
+        default="",
+        metavar="NODE",
+        help="publish the AMDGPUHealthy condition on this Node (in-cluster)",
+    )
     args = ap.parse_args(argv)
     try:
         agent = NodeAgent()
@@ -209,6 +258,23 @@ def main(argv=None) -> int:
     except NodeAgentError as e:
         print(json.dumps({"healthy": False, "agent_error": str(e)}))
         return 2
+    if args.patch_node:
+        import asyncio
+
+        from .kube.http import HTTPClient
+
+        async def publish():
+            kube = HTTPClient.from_service_account()
+            try:
+                await patch_node_condition(kube, args.patch_node, report)
+            finally:
+                await kube.close()
+
+        try:
+            asyncio.run(publish())
+        except Exception as e:
+            print(json.dumps({"healthy": report.healthy, "patch_error": str(e)}))
+            return 2
     out = asdict(report)
     if args.json:
         print(json.dumps(out, indent=2))
