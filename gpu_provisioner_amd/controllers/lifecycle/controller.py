@@ -399,19 +399,61 @@ class LifecycleController:
         ]
 
     async def _patch_status(self, nodeclaim: dict) -> None:
-        try:
-            updated = await self.kube.patch(
-                karpv1.API_VERSION,
-                karpv1.KIND_NODECLAIM,
-                ko.name_of(nodeclaim),
-                {"status": nodeclaim.get("status", {})},
-                subresource="status",
-            )
+        """Status write with optimistic locking: a merge patch of the whole
+        conditions list based on a stale read would silently erase conditions
+        written concurrently by the drift/termination controllers, so the
+        patch carries the in-hand resourceVersion and, on conflict, re-reads
+        and re-applies only the fields this controller owns."""
+        for _ in range(5):
+            try:
+                updated = await self.kube.patch(
+                    karpv1.API_VERSION,
+                    karpv1.KIND_NODECLAIM,
+                    ko.name_of(nodeclaim),
+                    {
+                        "metadata": {
+                            "resourceVersion": ko.meta(nodeclaim).get("resourceVersion")
+                        },
+                        "status": nodeclaim.get("status", {}),
+                    },
+                    subresource="status",
+                )
+            except ConflictError:
+                try:
+                    fresh = await self.kube.get(
+                        karpv1.API_VERSION, karpv1.KIND_NODECLAIM, ko.name_of(nodeclaim)
+                    )
+                except NotFoundError:
+                    return
+                # graft our status onto the fresh object: scalar fields we
+                # own replace, our conditions merge by type (foreign
+                # conditions like Drifted/Drained survive)
+                ours = nodeclaim.get("status", {})
+                fresh_status = fresh.setdefault("status", {})
+                for f in ("providerID", "imageID", "nodeName", "capacity", "allocatable"):
+                    if f in ours:
+                        fresh_status[f] = ours[f]
+                for cond in ours.get("conditions") or []:
+                    ko.set_condition(
+                        fresh,
+                        cond.get("type", ""),
+                        cond.get("status", ""),
+                        cond.get("reason", ""),
+                        cond.get("message", ""),
+                    )
+                nodeclaim["status"] = fresh_status
+                ko.meta(nodeclaim)["resourceVersion"] = fresh["metadata"]["resourceVersion"]
+                continue
+            except NotFoundError:
+                return
             # keep the in-hand object's resourceVersion fresh so a later
             # update (e.g. finalizer removal) doesn't conflict with our own write
             ko.meta(nodeclaim)["resourceVersion"] = updated["metadata"]["resourceVersion"]
-        except NotFoundError:
-            pass
+            return
+        log.warning(
+            "NodeClaim %s: status patch abandoned after repeated conflicts",
+            ko.name_of(nodeclaim),
+        )
 
     async def _delete_nodeclaim(self, nodeclaim: dict) -> None:
         try:

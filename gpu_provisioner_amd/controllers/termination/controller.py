@@ -315,24 +315,35 @@ class TerminationController:
     async def _set_nodeclaim_condition(
         self, nodeclaim: dict, cond: str, status: str, reason: str, message: str = ""
     ) -> bool:
-        """Patch one condition on the NodeClaim status; returns True if it
-        transitioned."""
-        try:
-            fresh = await self.kube.get(
-                karpv1.API_VERSION, karpv1.KIND_NODECLAIM, ko.name_of(nodeclaim)
-            )
-        except NotFoundError:
-            return False
-        changed = ko.set_condition(fresh, cond, status, reason, message)
-        if changed:
+        """Patch one condition on the NodeClaim status with optimistic
+        locking (a raced merge patch of the conditions list would clobber
+        concurrent lifecycle/drift writes); returns True if it transitioned."""
+        for _ in range(5):
+            try:
+                fresh = await self.kube.get(
+                    karpv1.API_VERSION, karpv1.KIND_NODECLAIM, ko.name_of(nodeclaim)
+                )
+            except NotFoundError:
+                return False
+            changed = ko.set_condition(fresh, cond, status, reason, message)
+            if not changed:
+                return False
             try:
                 await self.kube.patch(
                     karpv1.API_VERSION,
                     karpv1.KIND_NODECLAIM,
                     ko.name_of(fresh),
-                    {"status": {"conditions": fresh["status"]["conditions"]}},
+                    {
+                        "metadata": {
+                            "resourceVersion": fresh["metadata"].get("resourceVersion")
+                        },
+                        "status": {"conditions": fresh["status"]["conditions"]},
+                    },
                     subresource="status",
                 )
+            except ConflictError:
+                continue
             except NotFoundError:
                 return False
-        return changed
+            return True
+        return False

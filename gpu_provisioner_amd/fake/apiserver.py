@@ -224,11 +224,19 @@ class InMemoryAPIServer:
             cur = self._store[gvk].get(key)
             if cur is None:
                 raise NotFoundError(f"{kind} {namespace}/{name} not found")
+            # a merge patch that includes metadata.resourceVersion is an
+            # optimistic-lock precondition (real apiserver semantics);
+            # without it, merge wins unconditionally
+            req_rv = (patch.get("metadata") or {}).get("resourceVersion")
+            if req_rv and req_rv != cur["metadata"]["resourceVersion"]:
+                raise ConflictError(
+                    f"{kind} {name}: resourceVersion mismatch "
+                    f"({req_rv} != {cur['metadata']['resourceVersion']})"
+                )
             if subresource == "status":
                 merged_status = json_merge_patch(cur.get("status", {}), patch.get("status", patch))
                 return self._apply_update(gvk, key, cur, {"status": merged_status}, "status")
             merged = json_merge_patch(cur, patch)
-            # patches may not carry resourceVersion conflicts — merge wins
             merged.setdefault("metadata", {})["resourceVersion"] = cur["metadata"]["resourceVersion"]
             return self._apply_update(gvk, key, cur, merged, "")
 

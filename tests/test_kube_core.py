@@ -268,3 +268,58 @@ def test_apiserver_uid_precondition_delete():
             await c.get("karpenter.sh/v1", "NodeClaim", "a")
 
     run(main())
+
+
+def test_patch_with_resource_version_precondition():
+    """A merge patch carrying metadata.resourceVersion is an optimistic-lock
+    precondition (real apiserver semantics); without it, merge wins."""
+    from gpu_provisioner_amd.fake.apiserver import InMemoryAPIServer, InMemoryClient
+    from gpu_provisioner_amd.kube.client import ConflictError
+
+    async def main():
+        kube = InMemoryClient(InMemoryAPIServer())
+        await kube.create(
+            {"apiVersion": "v1", "kind": "Node", "metadata": {"name": "n1"},
+             "spec": {}, "status": {}}
+        )
+        n1 = await kube.get("v1", "Node", "n1")
+        stale_rv = n1["metadata"]["resourceVersion"]
+        await kube.patch("v1", "Node", "n1", {"metadata": {"labels": {"a": "1"}}})
+        with pytest.raises(ConflictError):
+            await kube.patch(
+                "v1", "Node", "n1",
+                {"metadata": {"resourceVersion": stale_rv, "labels": {"b": "2"}}},
+            )
+        # no precondition: merge wins over the stale read
+        got = await kube.patch("v1", "Node", "n1", {"metadata": {"labels": {"c": "3"}}})
+        assert got["metadata"]["labels"]["c"] == "3"
+
+    run(main())
+
+
+def test_lifecycle_status_patch_preserves_foreign_conditions():
+    """A condition written by another controller (e.g. Drifted) between the
+    lifecycle controller's read and its status patch must survive — the
+    conflict-retry path re-grafts only the fields lifecycle owns."""
+    from gpu_provisioner_amd.apis import v1 as karpv1
+    from gpu_provisioner_amd.fake.harness import Harness
+
+    async def main():
+        h = Harness().add_all_controllers(gc_interval=60.0, with_drift=False)
+        # don't start controllers: drive _patch_status by hand
+        nc = h.make_nodeclaim("race1")
+        await h.kube.create(nc)
+        in_hand = await h.kube.get(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "race1")
+        ko.set_condition(in_hand, karpv1.COND_LAUNCHED, ko.CONDITION_TRUE, "Launched")
+
+        # a foreign writer lands between lifecycle's read and its patch
+        foreign = await h.kube.get(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "race1")
+        ko.set_condition(foreign, karpv1.COND_DRIFTED, ko.CONDITION_TRUE, "NodeImageDrift")
+        await h.kube.update_status(foreign)
+
+        await h.lifecycle._patch_status(in_hand)
+        final = await h.kube.get(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "race1")
+        assert ko.condition_is_true(final, karpv1.COND_LAUNCHED)
+        assert ko.condition_is_true(final, karpv1.COND_DRIFTED), "foreign condition clobbered"
+
+    run(main())
