@@ -95,16 +95,31 @@ class Informer:
     def start(self) -> asyncio.Task:
         if self._task is None:
             self._task = asyncio.create_task(self._run(), name=f"informer-{self.kind}")
+            if self.resync_period > 0:
+                self._resync_task = asyncio.create_task(
+                    self._resync_loop(), name=f"informer-resync-{self.kind}"
+                )
         return self._task
 
     async def stop(self) -> None:
-        if self._task:
-            self._task.cancel()
-            try:
-                await self._task
-            except (asyncio.CancelledError, Exception):
-                pass
-            self._task = None
+        for attr in ("_task", "_resync_task"):
+            task = getattr(self, attr, None)
+            if task:
+                task.cancel()
+                try:
+                    await task
+                except (asyncio.CancelledError, Exception):
+                    pass
+                setattr(self, attr, None)
+
+    async def _resync_loop(self) -> None:
+        """client-go resync: periodically re-deliver every cached object as
+        MODIFIED so level-triggered controllers recover from any missed or
+        mishandled event."""
+        while True:
+            await asyncio.sleep(self.resync_period)
+            for obj in list(self._cache.values()):
+                self._notify(MODIFIED, obj)
 
     async def _run(self) -> None:
         backoff = 0.05

@@ -110,3 +110,26 @@ def test_informer_factory_shares_instances():
     b = f.informer("v1", "Node")
     c = f.informer("v1", "Pod")
     assert a is b and a is not c
+
+
+def test_resync_redelivers_cached_objects():
+    from gpu_provisioner_amd.fake.apiserver import InMemoryAPIServer, InMemoryClient
+    from gpu_provisioner_amd.kube.informer import Informer
+
+    async def main():
+        kube = InMemoryClient(InMemoryAPIServer())
+        await kube.create(
+            {"apiVersion": "v1", "kind": "Node", "metadata": {"name": "rs1"},
+             "spec": {}, "status": {}}
+        )
+        inf = Informer(kube, "v1", "Node", resync_period=0.05)
+        events = []
+        inf.add_handler(lambda et, obj: events.append((et, obj["metadata"]["name"])))
+        inf.start()
+        await inf.wait_for_sync()
+        await asyncio.sleep(0.18)  # ~3 resync periods
+        await inf.stop()
+        resyncs = [e for e in events if e == ("MODIFIED", "rs1")]
+        assert len(resyncs) >= 2, events
+
+    run(main())
