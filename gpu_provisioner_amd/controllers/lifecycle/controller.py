@@ -13,9 +13,12 @@ Design differences from the reference (deliberate):
   * reads go through the client (read-your-writes on the in-memory path and
     resourceVersion-fresh on HTTP), so the reference's 1 s post-patch sleep
     (controller.go:160-173) is unnecessary — provision p50 improves by ~1 s;
-  * the liveness sub-reconciler stays out (the reference disabled it,
-    controller.go:154, because agent-pool creates legitimately exceed the
-    15-min registration timeout).
+  * the liveness sub-reconciler (delete NodeClaims that never Register
+    within a timeout) is IMPLEMENTED but off by default: the reference
+    disabled it in-tree (controller.go:154) because agent-pool creates can
+    legitimately exceed the upstream 15-min registration timeout. Enable
+    with the RegistrationLiveness feature gate; the timeout here is 30 min
+    to leave room for ROCm driver installation on first boot.
 """
 from __future__ import annotations
 
@@ -53,6 +56,10 @@ log = logging.getLogger(__name__)
 INSTANCE_TERMINATION_REQUEUE = 5.0  # reference lifecycle/controller.go:241
 REGISTRATION_REQUEUE = 1.0
 LAUNCH_CACHE_TTL = 60.0
+# registration-liveness timeout (upstream karpenter uses 15 min; +15 for the
+# ROCm/amdgpu driver install on first boot). Only active when the
+# RegistrationLiveness gate is on — the reference ships it disabled.
+REGISTRATION_TTL_SECONDS = 30 * 60.0
 
 
 class LifecycleController:
@@ -67,8 +74,10 @@ class LifecycleController:
         nodes: Informer,
         workers: Optional[int] = None,
         termination_requeue: float = INSTANCE_TERMINATION_REQUEUE,
+        registration_ttl: Optional[float] = None,  # None = liveness disabled
     ):
         self.termination_requeue = termination_requeue
+        self.registration_ttl = registration_ttl
         self.kube = kube
         self.cloud = cloud
         self.recorder = recorder
@@ -129,13 +138,37 @@ class LifecycleController:
                 return Result(requeue=True)
 
         results = []
-        for sub in (self.launch, self.registration, self.initialization):
+        for sub in (self.launch, self.registration, self.initialization, self.liveness):
             res = await sub(nodeclaim)
             if res is not None:
                 results.append(res)
         requeues = [r.requeue_after for r in results if r.requeue_after is not None]
         if requeues:
             return Result(requeue_after=min(requeues))
+        return None
+
+    # -- liveness (registration timeout; gated, reference controller.go:154) --
+
+    async def liveness(self, nodeclaim: dict) -> Optional[Result]:
+        if self.registration_ttl is None or karpv1.is_registered(nodeclaim):
+            return None
+        created = ko.creation_timestamp_of(nodeclaim)
+        if created is None:
+            return None
+        elapsed = (ko.now() - created).total_seconds()
+        if elapsed < self.registration_ttl:
+            return Result(requeue_after=self.registration_ttl - elapsed)
+        log.warning(
+            "NodeClaim %s not Registered within %.0fs — deleting (RegistrationLiveness)",
+            ko.name_of(nodeclaim), self.registration_ttl,
+        )
+        self.recorder.publish(
+            nodeclaim,
+            "RegistrationTimeout",
+            f"node failed to register within {self.registration_ttl:.0f}s; deleting",
+            "Warning",
+        )
+        await self._delete_nodeclaim(nodeclaim)
         return None
 
     # -- launch ---------------------------------------------------------------

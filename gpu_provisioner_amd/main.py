@@ -46,10 +46,19 @@ def build_manager(kube, options: Options, cloud_provider, version: str = "0.1.0"
     volumeattachments = informers.informer("storage.k8s.io/v1", "VolumeAttachment")
     recorder = EventRecorder(kube)
 
+    from .controllers.lifecycle.controller import REGISTRATION_TTL_SECONDS
+
     eviction_queue = EvictionQueue(kube, recorder)
     controllers = [
         eviction_queue,
-        LifecycleController(kube, cloud, recorder, nodeclaims, nodes),
+        LifecycleController(
+            kube, cloud, recorder, nodeclaims, nodes,
+            registration_ttl=(
+                REGISTRATION_TTL_SECONDS
+                if options.feature_gates.registration_liveness
+                else None
+            ),
+        ),
         TerminationController(
             kube, cloud, recorder, nodes, nodeclaims, pods, volumeattachments, eviction_queue
         ),

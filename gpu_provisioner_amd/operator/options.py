@@ -23,6 +23,10 @@ class FeatureGates:
     # active disruption controllers).
     drift: bool = True
     drift_replace: bool = False
+    # delete NodeClaims that never reach Registered within the 30-min
+    # liveness window; off by default (the reference ships liveness
+    # disabled — AKS agent-pool creates can legitimately run long)
+    registration_liveness: bool = False
 
     @classmethod
     def parse(cls, s: str) -> "FeatureGates":
@@ -40,6 +44,8 @@ class FeatureGates:
                 gates.drift = enabled
             elif name == "DriftReplace":
                 gates.drift_replace = enabled
+            elif name == "RegistrationLiveness":
+                gates.registration_liveness = enabled
         return gates
 
 

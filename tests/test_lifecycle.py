@@ -278,3 +278,58 @@ def test_invalid_agent_pool_name_rejected():
             await h.stop()
 
     run(main())
+
+
+def test_registration_liveness_disabled_by_default():
+    """The reference ships liveness disabled (controller.go:154); ours is a
+    deliberate, gated choice — no TTL unless RegistrationLiveness is on."""
+    from gpu_provisioner_amd.fake.harness import Harness
+
+    h = Harness().add_all_controllers()
+    assert h.lifecycle.registration_ttl is None
+
+
+def test_registration_liveness_deletes_stuck_nodeclaim():
+    """With the gate on, a NodeClaim that launches but never Registers (node
+    never appears) is deleted once the TTL elapses."""
+    import gpu_provisioner_amd.controllers.lifecycle.controller as lc
+    from gpu_provisioner_amd.fake.harness import Harness
+
+    async def main():
+        # AKS sim that never brings the node up: infinite ready latency
+        h = Harness(ready_latency=3600.0, node_wait_interval=0.01)
+        h.instances.node_wait_attempts = 2  # launch fails fast on node wait
+        h.add_all_controllers(gc_interval=120.0, with_drift=False)
+        h.lifecycle.registration_ttl = 0.5
+        await h.start()
+        try:
+            await h.kube.create(h.make_nodeclaim("stuck1"))
+            # never reaches Registered; liveness deletes it after the TTL
+            await h.wait_gone(
+                karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "stuck1", timeout=30
+            )
+        finally:
+            await h.stop()
+
+    run(main())
+
+
+def test_registration_liveness_gate_wiring():
+    from gpu_provisioner_amd.controllers.lifecycle.controller import (
+        REGISTRATION_TTL_SECONDS,
+        LifecycleController,
+    )
+    from gpu_provisioner_amd.fake.harness import Harness
+    from gpu_provisioner_amd.main import build_manager
+    from gpu_provisioner_amd.operator.options import Options
+
+    h = Harness()
+    opts = Options.from_env_and_args(
+        ["--feature-gates", "RegistrationLiveness=true"], {}
+    )
+    mgr = build_manager(h.kube, opts, h.cloud.inner)
+    lc = [c for c in mgr.controllers if isinstance(c, LifecycleController)][0]
+    assert lc.registration_ttl == REGISTRATION_TTL_SECONDS
+    mgr2 = build_manager(h.kube, Options.from_env_and_args([], {}), h.cloud.inner)
+    lc2 = [c for c in mgr2.controllers if isinstance(c, LifecycleController)][0]
+    assert lc2.registration_ttl is None
