@@ -8,6 +8,7 @@ workload-identity webhook or federated credential is missing.
 from __future__ import annotations
 
 import logging
+import os
 
 from ..auth.config import AzureConfig, ConfigError, build_azure_config
 from ..auth.cred import new_credential
@@ -36,11 +37,19 @@ class Operator:
         except ConfigError as e:
             raise SystemExit(f"{e}\n{FEDERATED_CREDENTIAL_HINT}") from e
         self.credential = new_credential(self.config)
+        # E2E pipelines stamp scenario headers on every ARM request
+        # (reference azure_client.go:113-141 injects them via a pipeline
+        # policy when E2E_TEST_MODE is set)
+        env = environ if environ is not None else os.environ
+        extra_headers = {}
+        if env.get("E2E_TEST_MODE", "").lower() == "true":
+            extra_headers["X-Kaito-E2E"] = env.get("E2E_SCENARIO", "true")
         self.agent_pools = ARMAgentPoolsClient(
             self.credential,
             self.config.subscription_id,
             endpoint=self.config.arm_endpoint,
             user_agent=self.config.user_agent,
+            extra_headers=extra_headers,
         )
         self.catalog = InstanceTypeProvider(region=self.config.location)
         self.instances = InstanceProvider(

@@ -92,12 +92,16 @@ class ARMAgentPoolsClient(AgentPoolsAPI):
         http: Optional[httpx.AsyncClient] = None,
         lro_poll_interval: float = 5.0,
         max_retries: int = MAX_RETRIES,
+        extra_headers: Optional[dict] = None,
     ):
         self.credential = credential
         self.subscription_id = subscription_id
         self.endpoint = endpoint.rstrip("/")
         self.lro_poll_interval = lro_poll_interval
         self.max_retries = max_retries
+        # per-request header injection — the reference's E2E pipeline policy
+        # (azure_client.go:113-141) stamps test-scenario headers this way
+        self.extra_headers = dict(extra_headers or {})
         self.http = http or httpx.AsyncClient(
             timeout=httpx.Timeout(30.0, read=120.0),
             # the reference fans requests over an armbalancer pool of 100
@@ -132,7 +136,7 @@ class ARMAgentPoolsClient(AgentPoolsAPI):
                     url,
                     json=json_body,
                     params=params,
-                    headers={"Authorization": f"Bearer {token}"},
+                    headers={"Authorization": f"Bearer {token}", **self.extra_headers},
                 )
             except httpx.TransportError as e:
                 last_exc = e

@@ -163,3 +163,33 @@ def test_lro_failure_raises_with_code():
         assert exc.value.code == "SkuNotAvailable"
 
     run(main())
+
+
+def test_extra_headers_injected_on_every_request():
+    """E2E pipeline header injection (reference azure_client.go:113-141)."""
+    import httpx
+
+    from gpu_provisioner_amd.providers.instance.armclient import ARMAgentPoolsClient
+
+    seen = {}
+
+    async def handler(request: httpx.Request) -> httpx.Response:
+        seen.update(dict(request.headers))
+        return httpx.Response(200, json={"name": "p1", "properties": {}})
+
+    class FakeCred:
+        async def get_token(self):
+            return "tok"
+
+    async def main():
+        client = ARMAgentPoolsClient(
+            FakeCred(), "sub1",
+            http=httpx.AsyncClient(transport=httpx.MockTransport(handler)),
+            extra_headers={"X-Kaito-E2E": "scenario-7"},
+        )
+        await client.get("rg", "cluster", "p1")
+        assert seen.get("x-kaito-e2e") == "scenario-7"
+        assert seen.get("authorization") == "Bearer tok"
+        await client.close()
+
+    run(main())
