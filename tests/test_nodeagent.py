@@ -75,8 +75,8 @@ def test_gpu_hbm_bandwidth_floor():
     _ensure_lib()
     agent = NodeAgent()
     bw = agent.hbm_bandwidth(0, bytes_=1 << 30, iters=10)
-    # float4 streaming copy reaches ~6 TB/s on healthy MI355X; 4 TB/s floor
-    assert bw > 4000.0, f"HBM bandwidth {bw:.0f} GB/s below healthy floor"
+    # contiguous-chunk nt copy measures ~5.7 TB/s on healthy MI355X
+    assert bw > 4800.0, f"HBM bandwidth {bw:.0f} GB/s below healthy floor"
 
 
 @pytest.mark.gpu
