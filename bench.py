@@ -35,10 +35,14 @@ from gpu_provisioner_amd.fake.harness import Harness  # noqa: E402
 VM_SIZE = "Standard_ND128isr_MI355X_v6"
 
 
-def build_harness() -> Harness:
+def build_harness(create_latency: float = 0.0, ready_latency: float = 0.0) -> Harness:
     # the full production controller topology: lifecycle + termination +
     # eviction + both GCs (health excluded: no repairs during a clean bench)
-    return Harness(node_wait_interval=0.002).add_all_controllers(
+    return Harness(
+        create_latency=create_latency,
+        ready_latency=ready_latency,
+        node_wait_interval=0.002,
+    ).add_all_controllers(
         lifecycle_workers=256,
         termination_requeue=0.002,
         drain_requeue=0.002,
@@ -66,8 +70,11 @@ async def one_step(h: Harness, step: int, concurrent: int, latencies: list) -> N
     await asyncio.gather(*(teardown(n) for n in names))
 
 
-async def run_bench(steps: int, warmup: int, concurrent: int) -> dict:
-    h = build_harness()
+async def run_bench(
+    steps: int, warmup: int, concurrent: int,
+    create_latency: float = 0.0, ready_latency: float = 0.0,
+) -> dict:
+    h = build_harness(create_latency, ready_latency)
     await h.start()
     try:
         warm_lat: list = []
@@ -90,6 +97,10 @@ def main() -> None:
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--concurrent", type=int, default=8,
                     help="NodeClaims provisioned per step (8 = one full MI355X host worth)")
+    ap.add_argument("--create-latency", type=float, default=0.0,
+                    help="simulated agent-pool LRO seconds (0 = BASELINE config #1)")
+    ap.add_argument("--ready-latency", type=float, default=0.0,
+                    help="simulated node boot-to-Ready seconds")
     args = ap.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -115,7 +126,12 @@ def main() -> None:
             torch.cuda.synchronize()
 
     sync()
-    result = asyncio.run(run_bench(args.steps, args.warmup, args.concurrent))
+    result = asyncio.run(
+        run_bench(
+            args.steps, args.warmup, args.concurrent,
+            args.create_latency, args.ready_latency,
+        )
+    )
     sync()
 
     elapsed = result["elapsed_s"]
@@ -161,7 +177,14 @@ def main() -> None:
                         "model": "MI355X NodeClaim provisioning (karpenter CloudProvider contract)",
                         "vm_sku": VM_SIZE,
                         "concurrent_nodeclaims": args.concurrent,
-                        "cloud": "in-memory fake AKS, zero latency (BASELINE config #1)",
+                        "cloud": (
+                            "in-memory fake AKS, zero latency (BASELINE config #1)"
+                            if not (args.create_latency or args.ready_latency)
+                            else (
+                                f"in-memory fake AKS, simulated LRO {args.create_latency}s"
+                                f" + node-ready {args.ready_latency}s"
+                            )
+                        ),
                         "cycle": "create→Launched→Registered→Initialized(amd.com/gpu)→delete→gone",
                         "p50_ready_latency_s": round(p50, 4),
                         "p95_ready_latency_s": round(p95, 4),
