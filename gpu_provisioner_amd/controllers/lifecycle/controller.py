@@ -276,6 +276,11 @@ class LifecycleController:
             "metadata": {
                 "labels": node_labels,
                 "finalizers": node_finalizers,
+                # optimistic lock: the taint write below is merged from OUR
+                # read of the node — without the precondition it can race
+                # the kubelet's one-shot removal of the not-ready startup
+                # taint and resurrect it forever (found by the chaos test)
+                "resourceVersion": ko.meta(node).get("resourceVersion"),
                 "ownerReferences": [
                     {
                         "apiVersion": karpv1.API_VERSION,
@@ -288,7 +293,12 @@ class LifecycleController:
             },
             "spec": {"taints": node_taints or None},
         }
-        await self.kube.patch("v1", "Node", ko.name_of(node), patch)
+        try:
+            await self.kube.patch("v1", "Node", ko.name_of(node), patch)
+        except ConflictError:
+            # node changed under us (e.g. kubelet stripped a startup taint):
+            # re-run registration against the fresh node
+            return Result(requeue_after=REGISTRATION_REQUEUE)
         status = nodeclaim.setdefault("status", {})
         status["nodeName"] = ko.name_of(node)
         ko.set_condition(nodeclaim, karpv1.COND_REGISTERED, ko.CONDITION_TRUE, "Registered")

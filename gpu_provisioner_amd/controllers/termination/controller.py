@@ -253,7 +253,13 @@ class TerminationController:
         needs_label = labels.get(karpv1.EXCLUDE_FROM_LB_LABEL_KEY) != "karpenter"
         if not needs_taint and not needs_label:
             return
-        patch: dict = {"metadata": {}, "spec": {}}
+        # optimistic lock on the taint merge: same lost-update race as
+        # registration's taint sync (ConflictError surfaces to reconcile,
+        # which requeues against a fresh read)
+        patch: dict = {
+            "metadata": {"resourceVersion": ko.meta(node).get("resourceVersion")},
+            "spec": {},
+        }
         if needs_taint:
             patch["spec"]["taints"] = ko.merge_taints(taints, [desired])
         if needs_label:

@@ -1,0 +1,85 @@
+"""Chaos convergence: random retryable ARM failures on every cloud verb
+while a 24-claim fleet provisions and tears down. Level-triggered
+controllers with rate-limited requeues must converge despite the noise —
+the distributed-systems property the reference delegates to live-cluster
+e2e, exercised here in-process with a seeded RNG."""
+import asyncio
+import random
+
+from gpu_provisioner_amd.apis import v1 as karpv1
+from gpu_provisioner_amd.fake.harness import Harness
+from gpu_provisioner_amd.providers.instance.armapi import ARMError
+from tests.conftest import run
+
+FLEET = 24
+FAIL_P = 0.15
+
+
+class ChaosError:
+    """Drop-in for fake.agentpools.ScriptedError: raises a retryable ARM
+    error with probability p (seeded — the test is deterministic)."""
+
+    def __init__(self, rng: random.Random, p: float = FAIL_P):
+        self.rng = rng
+        self.p = p
+        self.raised = 0
+
+    def check(self) -> None:
+        if self.rng.random() < self.p:
+            self.raised += 1
+            raise ARMError(503, "ServerBusy", "chaos: simulated ARM brownout")
+
+
+def test_fleet_converges_under_arm_chaos():
+    async def main():
+        rng = random.Random(20260913)
+        h = Harness(node_wait_interval=0.005).add_all_controllers(
+            lifecycle_workers=64,
+            termination_requeue=0.01,
+            drain_requeue=0.01,
+            instance_poll=0.01,
+            gc_interval=1.0,
+            adoption_age=0.5,
+        )
+        chaos = [ChaosError(rng) for _ in range(4)]
+        (
+            h.agent_pools.create_error,
+            h.agent_pools.delete_error,
+            h.agent_pools.get_error,
+            h.agent_pools.list_error,
+        ) = chaos
+        await h.start()
+        try:
+            names = [f"chaos{i:02d}" for i in range(FLEET)]
+            await asyncio.gather(
+                *(h.kube.create(h.make_nodeclaim(n)) for n in names)
+            )
+            done = await asyncio.gather(
+                *(h.wait_initialized(n, timeout=60) for n in names)
+            )
+            assert all(karpv1.is_initialized(nc) for nc in done)
+            assert sum(c.raised for c in chaos) > 0, "chaos never fired — test is vacuous"
+
+            await asyncio.gather(
+                *(
+                    h.kube.delete(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, n)
+                    for n in names
+                )
+            )
+            await asyncio.gather(
+                *(
+                    h.wait_gone(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, n, timeout=60)
+                    for n in names
+                )
+            )
+
+            async def pools_empty():
+                return not h.agent_pools.pools or None
+
+            await h.wait_for(pools_empty, timeout=60)
+            # no leaked nodes either
+            assert await h.kube.list("v1", "Node") == []
+        finally:
+            await h.stop()
+
+    run(main(), timeout=240)
