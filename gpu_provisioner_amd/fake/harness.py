@@ -142,6 +142,22 @@ class Harness:
             self.controllers.append(self.drift)
         return self
 
+    async def crash_restart_controllers(self, **controller_kwargs) -> None:
+        """Simulate a controller-manager crash+restart: stop every
+        controller, discard ALL in-memory controller state (workqueues,
+        launch idempotency caches, eviction dedup) and start fresh
+        instances against the same apiserver/cloud state. Informers re-sync
+        from the server as a restarted manager's caches would."""
+        for c in self.controllers:
+            await c.controller.stop()
+        self.controllers = []
+        # drop the dead controllers' event handlers (their queues are shut)
+        for inf in self.informers._informers.values():
+            inf._handlers.clear()
+        self.add_all_controllers(**controller_kwargs)
+        for c in self.controllers:
+            c.controller.start()
+
     # -- lifecycle -----------------------------------------------------------
 
     async def start(self) -> None:

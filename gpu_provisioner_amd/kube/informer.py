@@ -61,8 +61,18 @@ class Informer:
         return name in self._indexes
 
     def add_handler(self, fn: Callable) -> None:
-        """fn(event_type, obj) — called for ADDED/MODIFIED/DELETED."""
+        """fn(event_type, obj) — called for ADDED/MODIFIED/DELETED.
+
+        client-go semantics: the current store is replayed to the new
+        handler as synthetic ADDED events, so a handler registered after
+        sync (e.g. a restarted controller) still sees the whole world —
+        level-triggered recovery depends on this."""
         self._handlers.append(fn)
+        for obj in list(self._cache.values()):
+            try:
+                fn(ADDED, obj)
+            except Exception:
+                log.exception("informer %s handler failed during replay", self.kind)
 
     # -- cache access -------------------------------------------------------
     #
