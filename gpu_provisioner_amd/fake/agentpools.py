@@ -20,7 +20,12 @@ from typing import AsyncIterator, Optional
 
 from ..apis import v1 as karpv1
 from ..kube import objects as ko
-from ..kube.client import KubeClient, NotFoundError
+from ..kube.client import (
+    AlreadyExistsError,
+    APIError,
+    KubeClient,
+    NotFoundError,
+)
 from ..providers.instance.armapi import (
     AgentPoolsAPI,
     ARMError,
@@ -248,10 +253,17 @@ class AKSSimulator:
                 },
             },
         }
-        try:
-            await self.kube.create(node)
-        except Exception as e:
-            log.debug("node %s create: %s", self.node_name(name), e)
+        # kubelet semantics: node registration retries until it lands
+        for _ in range(200):
+            try:
+                await self.kube.create(node)
+                break
+            except AlreadyExistsError:
+                break
+            except APIError:
+                await asyncio.sleep(0.02)
+        else:
+            log.warning("node %s never registered", self.node_name(name))
             return
         self._spawn(self._become_ready(name, props))
 
@@ -259,22 +271,43 @@ class AKSSimulator:
         if self.ready_latency:
             await asyncio.sleep(self.ready_latency)
         node_name = self.node_name(pool)
-        # kubelet Ready + drop the not-ready taint
-        try:
-            node = await self.kube.get("v1", "Node", node_name)
-        except NotFoundError:
-            return
-        taints = [
-            t
-            for t in node.get("spec", {}).get("taints") or []
-            if t.get("key") != "node.kubernetes.io/not-ready"
-        ]
-        await self.kube.patch(
-            "v1", "Node", node_name, {"spec": {"taints": taints or None}}
-        )
-        await self.kube.patch(
-            "v1",
-            "Node",
+        # kubelet Ready + drop the not-ready taint. The removal is
+        # read-modify-write on spec.taints, so it carries the node's
+        # resourceVersion as an optimistic lock (a raced unconditioned merge
+        # would clobber the registration controller's taint sync — the same
+        # lost-update bug in the other direction) and retries like a real
+        # kubelet until it lands.
+        for _ in range(200):
+            try:
+                node = await self.kube.get("v1", "Node", node_name)
+            except NotFoundError:
+                return
+            except APIError:
+                await asyncio.sleep(0.02)
+                continue
+            taints = [
+                t
+                for t in node.get("spec", {}).get("taints") or []
+                if t.get("key") != "node.kubernetes.io/not-ready"
+            ]
+            try:
+                await self.kube.patch(
+                    "v1",
+                    "Node",
+                    node_name,
+                    {
+                        "metadata": {
+                            "resourceVersion": node["metadata"].get("resourceVersion")
+                        },
+                        "spec": {"taints": taints or None},
+                    },
+                )
+                break
+            except NotFoundError:
+                return
+            except APIError:
+                await asyncio.sleep(0.02)
+        await self._retry_status_patch(
             node_name,
             {
                 "status": {
@@ -283,15 +316,12 @@ class AKSSimulator:
                     ]
                 }
             },
-            subresource="status",
         )
         if self.plugin_latency:
             await asyncio.sleep(self.plugin_latency)
         # AMD device plugin registers amd.com/gpu
         gpus = str(self.gpu_count_for(props.get("vmSize", "")))
-        await self.kube.patch(
-            "v1",
-            "Node",
+        await self._retry_status_patch(
             node_name,
             {
                 "status": {
@@ -299,14 +329,27 @@ class AKSSimulator:
                     "allocatable": {karpv1.AMD_GPU_RESOURCE: gpus},
                 }
             },
-            subresource="status",
         )
 
+    async def _retry_status_patch(self, node_name: str, patch: dict) -> None:
+        for _ in range(200):
+            try:
+                await self.kube.patch("v1", "Node", node_name, patch, subresource="status")
+                return
+            except NotFoundError:
+                return
+            except APIError:
+                await asyncio.sleep(0.02)
+
     async def _on_pool_deleted(self, pool: str) -> None:
-        try:
-            await self.kube.delete("v1", "Node", self.node_name(pool))
-        except NotFoundError:
-            pass
+        for _ in range(200):
+            try:
+                await self.kube.delete("v1", "Node", self.node_name(pool))
+                return
+            except NotFoundError:
+                return
+            except APIError:
+                await asyncio.sleep(0.02)
 
     def _spawn(self, coro) -> None:
         task = asyncio.get_event_loop().create_task(coro)

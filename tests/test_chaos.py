@@ -83,3 +83,79 @@ def test_fleet_converges_under_arm_chaos():
             await h.stop()
 
     run(main(), timeout=240)
+
+
+def test_fleet_converges_under_kube_and_arm_chaos():
+    """Both backends misbehave at once: every ARM verb AND every kube write
+    verb (update/patch/delete) fails randomly with retryable errors. The
+    stack must still converge a 16-claim fleet through provision+teardown."""
+    from gpu_provisioner_amd.kube.client import ConflictError, TooManyRequestsError
+
+    async def main():
+        rng = random.Random(777)
+        h = Harness(node_wait_interval=0.005).add_all_controllers(
+            lifecycle_workers=64,
+            termination_requeue=0.01,
+            drain_requeue=0.01,
+            instance_poll=0.01,
+            gc_interval=1.0,
+            adoption_age=0.5,
+        )
+        arm = [ChaosError(rng, p=0.10) for _ in range(4)]
+        (
+            h.agent_pools.create_error,
+            h.agent_pools.delete_error,
+            h.agent_pools.get_error,
+            h.agent_pools.list_error,
+        ) = arm
+        kube_fired = {"n": 0}
+
+        def kube_chaos(verb, gvk, payload):
+            # eviction uses its own path; pods excluded so drain's PDB retry
+            # semantics stay deterministic here
+            if verb in ("update", "patch", "delete") and gvk[1] != "Pod":
+                if rng.random() < 0.08:
+                    kube_fired["n"] += 1
+                    return (
+                        ConflictError("chaos: conflict")
+                        if rng.random() < 0.5
+                        else TooManyRequestsError("chaos: throttled")
+                    )
+            return None
+
+        h.server.reactors.append(kube_chaos)
+        await h.start()
+        try:
+            names = [f"kchaos{i:02d}" for i in range(16)]
+            await asyncio.gather(*(h.kube.create(h.make_nodeclaim(n)) for n in names))
+            done = await asyncio.gather(
+                *(h.wait_initialized(n, timeout=90) for n in names)
+            )
+            assert all(karpv1.is_initialized(nc) for nc in done)
+            assert kube_fired["n"] > 0 and sum(c.raised for c in arm) > 0
+
+            async def chaos_tolerant_delete(n):
+                # the test's own client retries like any well-behaved caller
+                for _ in range(50):
+                    try:
+                        await h.kube.delete(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, n)
+                        return
+                    except Exception:
+                        await asyncio.sleep(0.02)
+
+            await asyncio.gather(*(chaos_tolerant_delete(n) for n in names))
+            await asyncio.gather(
+                *(
+                    h.wait_gone(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, n, timeout=90)
+                    for n in names
+                )
+            )
+
+            async def pools_empty():
+                return not h.agent_pools.pools or None
+
+            await h.wait_for(pools_empty, timeout=90)
+        finally:
+            await h.stop()
+
+    run(main(), timeout=300)
