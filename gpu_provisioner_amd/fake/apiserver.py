@@ -43,6 +43,7 @@ from ..kube.client import (
 )
 
 _WATCH_HISTORY = 4096  # events kept for resourceVersion resume
+_WATCH_BROKEN = "__WATCH_BROKEN__"  # sentinel: simulated dropped stream
 
 
 def _key(namespace: str, name: str) -> tuple:
@@ -321,6 +322,16 @@ class InMemoryAPIServer:
 
         return queue, unsubscribe
 
+    def break_watches(self) -> int:
+        """Chaos hook: simulate every open watch stream dropping (as a real
+        apiserver does on timeout/netsplit). Watchers receive a stream error
+        and must relist+rewatch. Returns the number of streams broken."""
+        broken = 0
+        for _, queue in list(self._watchers):
+            queue.put_nowait((_WATCH_BROKEN, None))
+            broken += 1
+        return broken
+
 
 class InMemoryClient(KubeClient):
     """KubeClient over an InMemoryAPIServer."""
@@ -398,6 +409,8 @@ class InMemoryClient(KubeClient):
         try:
             while True:
                 event_type, obj = await queue.get()
+                if event_type == _WATCH_BROKEN:
+                    raise APIError("watch stream broken")
                 if namespace and ko.namespace_of(obj) != namespace:
                     continue
                 if sel and not sel.matches(ko.labels_of(obj)):

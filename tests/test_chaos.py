@@ -159,3 +159,53 @@ def test_fleet_converges_under_kube_and_arm_chaos():
             await h.stop()
 
     run(main(), timeout=300)
+
+
+def test_fleet_converges_while_watch_streams_drop():
+    """Every open watch stream drops repeatedly mid-provision (apiserver
+    timeout / netsplit shape): informers must relist+rewatch and the fleet
+    must still converge with no duplicate side effects."""
+
+    async def main():
+        h = Harness(node_wait_interval=0.005).add_all_controllers(
+            lifecycle_workers=64,
+            termination_requeue=0.01,
+            drain_requeue=0.01,
+            instance_poll=0.01,
+            gc_interval=1.0,
+            adoption_age=0.5,
+        )
+        await h.start()
+        try:
+            names = [f"wchaos{i:02d}" for i in range(12)]
+            await asyncio.gather(*(h.kube.create(h.make_nodeclaim(n)) for n in names))
+            broken = 0
+            for _ in range(5):
+                await asyncio.sleep(0.05)
+                broken += h.server.break_watches()
+            assert broken > 0
+            done = await asyncio.gather(
+                *(h.wait_initialized(n, timeout=60) for n in names)
+            )
+            assert all(karpv1.is_initialized(nc) for nc in done)
+            # exactly one pool and one create per claim despite the relists
+            assert sorted(h.agent_pools.pools) == sorted(names)
+            assert h.agent_pools.create_calls == len(names)
+
+            await asyncio.gather(
+                *(
+                    h.kube.delete(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, n)
+                    for n in names
+                )
+            )
+            h.server.break_watches()
+            await asyncio.gather(
+                *(
+                    h.wait_gone(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, n, timeout=60)
+                    for n in names
+                )
+            )
+        finally:
+            await h.stop()
+
+    run(main(), timeout=240)
