@@ -51,6 +51,12 @@ class HealthController:
         nodes.add_handler(self._on_node_event)
 
     def _on_node_event(self, event_type: str, obj: dict) -> None:
+        if event_type == "DELETED":
+            # drop first-seen bookkeeping for vanished nodes (slow leak)
+            name = ko.name_of(obj)
+            for k in [k for k in self._first_seen if k[0] == name]:
+                del self._first_seen[k]
+            return
         if karpv1.node_is_managed(obj):
             self.controller.enqueue_nowait(ko.name_of(obj))
 

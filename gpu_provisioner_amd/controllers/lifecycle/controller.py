@@ -203,6 +203,12 @@ class LifecycleController:
                 await self._patch_status(nodeclaim)
                 raise  # rate-limited retry via the workqueue
             self._launch_cache[uid] = (time.monotonic() + LAUNCH_CACHE_TTL, created)
+            # bound the cache: expired entries would otherwise accumulate
+            # forever under claim churn (a slow leak in a long-lived manager)
+            if len(self._launch_cache) > 128:
+                nw = time.monotonic()
+                for k in [k for k, v in self._launch_cache.items() if v[0] <= nw]:
+                    del self._launch_cache[k]
 
         # populate from the created instance (reference launch.go:126-140)
         labels = ko.labels_of(created)

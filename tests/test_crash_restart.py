@@ -125,3 +125,48 @@ def test_restart_gc_adopts_pool_leaked_while_down():
             await h.stop()
 
     run(main())
+
+
+def test_long_running_state_is_bounded_under_churn():
+    """A long-lived manager must not leak per-claim bookkeeping: after
+    churning 150 uniquely-named claims through provision+teardown, the
+    launch idempotency cache, backoff accounting and eviction bookkeeping
+    stay bounded."""
+    import gpu_provisioner_amd.controllers.lifecycle.controller as lc
+
+    async def main():
+        h = Harness(node_wait_interval=0.005).add_all_controllers(
+            gc_interval=30.0, termination_requeue=0.005, drain_requeue=0.005,
+            instance_poll=0.005,
+        )
+        # expire launch-cache entries almost immediately so the churn
+        # exercises the pruning path (TTL is module-level, restore after)
+        old_ttl = lc.LAUNCH_CACHE_TTL
+        lc.LAUNCH_CACHE_TTL = 0.01
+        await h.start()
+        try:
+            for batch in range(15):
+                names = [f"lk{batch:02d}x{i}" for i in range(10)]
+                await asyncio.gather(*(h.kube.create(h.make_nodeclaim(n)) for n in names))
+                await asyncio.gather(*(h.wait_initialized(n, timeout=30) for n in names))
+                await asyncio.gather(
+                    *(
+                        h.kube.delete(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, n)
+                        for n in names
+                    )
+                )
+                await asyncio.gather(
+                    *(
+                        h.wait_gone(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, n, timeout=30)
+                        for n in names
+                    )
+                )
+            assert len(h.lifecycle._launch_cache) <= 129, len(h.lifecycle._launch_cache)
+            assert len(h.eviction_queue._grace) == 0
+            # per-item backoff accounting is forgotten on successful reconcile
+            assert len(h.lifecycle.controller.queue.rate_limiter.backoff._failures) < 50
+        finally:
+            lc.LAUNCH_CACHE_TTL = old_ttl
+            await h.stop()
+
+    run(main(), timeout=240)
