@@ -38,15 +38,18 @@ VM_SIZE = "Standard_ND128isr_MI355X_v6"
 def build_harness(create_latency: float = 0.0, ready_latency: float = 0.0) -> Harness:
     # the full production controller topology: lifecycle + termination +
     # eviction + both GCs (health excluded: no repairs during a clean bench)
+    # progress is event-driven (node/pod/VA events re-trigger reconciles);
+    # the requeue intervals are BACKSTOPS. Sub-10ms backstops busy-poll and
+    # livelock the loop at high concurrency (128+ claims).
     return Harness(
         create_latency=create_latency,
         ready_latency=ready_latency,
-        node_wait_interval=0.002,
+        node_wait_interval=0.01,
     ).add_all_controllers(
         lifecycle_workers=256,
-        termination_requeue=0.002,
-        drain_requeue=0.002,
-        instance_poll=0.002,
+        termination_requeue=0.02,
+        drain_requeue=0.02,
+        instance_poll=0.02,
         gc_interval=30.0,
         with_health=False,
     )

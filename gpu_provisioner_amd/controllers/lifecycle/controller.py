@@ -252,14 +252,17 @@ class LifecycleController:
             return Result(requeue_after=REGISTRATION_REQUEUE)
         node = await self._node_by_provider_id(provider_id)
         if node is None:
-            ko.set_condition(
+            # patch only on TRANSITION: an unconditional write here turns the
+            # 1 s requeue into a per-claim event storm (every patch
+            # broadcasts → handlers re-enqueue → reconcile → patch …)
+            if ko.set_condition(
                 nodeclaim,
                 karpv1.COND_REGISTERED,
                 ko.CONDITION_FALSE,
                 "NodeNotFound",
                 "node has not registered with the cluster yet",
-            )
-            await self._patch_status(nodeclaim)
+            ):
+                await self._patch_status(nodeclaim)
             return Result(requeue_after=REGISTRATION_REQUEUE)
 
         # sync taints/labels/owner-ref onto the node (registration.go:117-147)
@@ -325,10 +328,10 @@ class LifecycleController:
             return Result(requeue_after=REGISTRATION_REQUEUE)
         reason = self._initialization_gate(nodeclaim, node)
         if reason:
-            ko.set_condition(
+            if ko.set_condition(
                 nodeclaim, karpv1.COND_INITIALIZED, ko.CONDITION_FALSE, "NotInitialized", reason
-            )
-            await self._patch_status(nodeclaim)
+            ):
+                await self._patch_status(nodeclaim)
             return Result(requeue_after=REGISTRATION_REQUEUE)
         await self.kube.patch(
             "v1",
@@ -398,10 +401,10 @@ class LifecycleController:
                         except NotFoundError:
                             pass
                 return Result(requeue_after=self.termination_requeue)
-        ko.set_condition(
+        if ko.set_condition(
             nodeclaim, karpv1.COND_INSTANCE_TERMINATING, ko.CONDITION_TRUE, "InstanceTerminating"
-        )
-        await self._patch_status(nodeclaim)
+        ):
+            await self._patch_status(nodeclaim)
         try:
             await self.cloud.delete(nodeclaim)
             return Result(requeue_after=self.termination_requeue)

@@ -91,10 +91,33 @@ class TerminationController:
                 "nodeName", lambda o: o.get("spec", {}).get("nodeName") or None
             )
         nodes.add_handler(self._on_node_event)
+        # drain progress is event-driven: pod/volume-attachment changes on a
+        # deleting node re-trigger its reconcile immediately, so the
+        # drain/volume requeue intervals are only backstops — without these
+        # handlers the controller busy-polls, which livelocks at high
+        # concurrency (measured: 128-claim teardown saturated the loop)
+        pods.add_handler(self._on_pod_event)
+        if volumeattachments is not None:
+            volumeattachments.add_handler(self._on_va_event)
 
     def _on_node_event(self, event_type: str, obj: dict) -> None:
         if ko.is_deleting(obj) and ko.has_finalizer(obj, karpv1.TERMINATION_FINALIZER):
             self.controller.enqueue_nowait(ko.name_of(obj))
+
+    def _on_pod_event(self, event_type: str, obj: dict) -> None:
+        node_name = obj.get("spec", {}).get("nodeName")
+        if node_name:
+            self._enqueue_if_terminating(node_name)
+
+    def _on_va_event(self, event_type: str, obj: dict) -> None:
+        node_name = obj.get("spec", {}).get("nodeName")
+        if node_name:
+            self._enqueue_if_terminating(node_name)
+
+    def _enqueue_if_terminating(self, node_name: str) -> None:
+        node = self.nodes.get(node_name) if self.nodes.has_synced else None
+        if node is not None and ko.is_deleting(node):
+            self.controller.enqueue_nowait(node_name)
 
     # ------------------------------------------------------------------ main
 

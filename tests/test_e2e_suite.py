@@ -337,3 +337,49 @@ def test_spec9_workload_pod_binds_to_provisioned_node():
             await h.stop()
 
     run(main())
+
+
+def test_spec12_128_concurrent_churn_completes():
+    """Regression for the teardown livelock: 128 concurrent NodeClaims
+    through a full churn cycle must complete — sub-10ms requeue backstops
+    plus unconditional condition patches once turned finalize into a
+    406k-reconciles-per-35s event storm that starved all progress."""
+
+    async def main():
+        import time
+
+        h = Harness(node_wait_interval=0.01).add_all_controllers(
+            lifecycle_workers=256, termination_requeue=0.02, drain_requeue=0.02,
+            instance_poll=0.02, gc_interval=60.0, with_health=False,
+        )
+        await h.start()
+        try:
+            names = [f"big{i:03d}" for i in range(128)]
+            t0 = time.monotonic()
+            await asyncio.gather(
+                *(
+                    h.kube.create(
+                        spec_nodeclaim(n, {karpv1.KAITO_WORKSPACE_LABEL_KEY: "w"})
+                    )
+                    for n in names
+                )
+            )
+            await asyncio.gather(*(h.wait_initialized(n, timeout=60) for n in names))
+            await asyncio.gather(
+                *(
+                    h.kube.delete(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, n)
+                    for n in names
+                )
+            )
+            await asyncio.gather(
+                *(
+                    h.wait_gone(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, n, timeout=60)
+                    for n in names
+                )
+            )
+            assert not h.agent_pools.pools
+            assert time.monotonic() - t0 < 90
+        finally:
+            await h.stop()
+
+    run(main(), timeout=200)
