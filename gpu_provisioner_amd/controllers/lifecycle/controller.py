@@ -400,7 +400,11 @@ class LifecycleController:
                             await self.kube.delete("v1", "Node", ko.name_of(node))
                         except NotFoundError:
                             pass
-                return Result(requeue_after=self.termination_requeue)
+                # node deletion completion is EVENT-driven (the node DELETED
+                # event maps back to this claim); this requeue is only a
+                # backstop. A short interval here multiplies into a
+                # fleet-wide polling storm while termination drains.
+                return Result(requeue_after=max(self.termination_requeue, 1.0))
         if ko.set_condition(
             nodeclaim, karpv1.COND_INSTANCE_TERMINATING, ko.CONDITION_TRUE, "InstanceTerminating"
         ):
