@@ -73,6 +73,7 @@ class Harness:
         self.informers = InformerFactory(self.kube)
         self.nodeclaims: Informer = self.informers.informer(karpv1.API_VERSION, karpv1.KIND_NODECLAIM)
         self.nodes: Informer = self.informers.informer("v1", "Node")
+        self.instances.set_nodes_informer(self.nodes)
         self.pods: Informer = self.informers.informer("v1", "Pod")
         self.pods.add_index("nodeName", lambda o: o.get("spec", {}).get("nodeName") or None)
         self.volumeattachments: Informer = self.informers.informer(

@@ -42,6 +42,10 @@ def build_manager(kube, options: Options, cloud_provider, version: str = "0.1.0"
     informers = InformerFactory(kube)
     nodeclaims = informers.informer(karpv1.API_VERSION, karpv1.KIND_NODECLAIM)
     nodes = informers.informer("v1", "Node")
+    # cache-backed node lookups for the instance provider's node wait
+    inner = getattr(cloud_provider, "instances", None)
+    if inner is not None and hasattr(inner, "set_nodes_informer"):
+        inner.set_nodes_informer(nodes)
     pods = informers.informer("v1", "Pod")
     volumeattachments = informers.informer("storage.k8s.io/v1", "VolumeAttachment")
     recorder = EventRecorder(kube)

@@ -89,6 +89,27 @@ class InstanceProvider:
         self.cluster_name = cluster_name
         self.node_wait_attempts = node_wait_attempts
         self.node_wait_interval = node_wait_interval
+        # optional Node informer for cache-backed lookups (the reference
+        # reads through controller-runtime's cached client; an apiserver
+        # LIST per wait attempt is O(cluster) and shows at 128 concurrent)
+        self.nodes_informer = None
+
+    def set_nodes_informer(self, informer) -> None:
+        informer.add_index(
+            "agentpool",
+            lambda o: [
+                v
+                for v in (
+                    (o.get("metadata", {}).get("labels") or {}).get(karpv1.AGENTPOOL_LABEL_KEY),
+                    (o.get("metadata", {}).get("labels") or {}).get(
+                        karpv1.AZURE_AGENTPOOL_LABEL_KEY
+                    ),
+                )
+                if v
+            ]
+            or None,
+        )
+        self.nodes_informer = informer
 
     # ------------------------------------------------------------------ create
 
@@ -198,6 +219,10 @@ class InstanceProvider:
         return "", None
 
     async def _node_for_pool(self, pool: str) -> Optional[dict]:
+        inf = self.nodes_informer
+        if inf is not None and inf.has_synced:
+            nodes = inf.by_index("agentpool", pool)
+            return nodes[0] if nodes else None
         for selector in (
             f"{karpv1.AGENTPOOL_LABEL_KEY}={pool}",
             f"{karpv1.AZURE_AGENTPOOL_LABEL_KEY}={pool}",
