@@ -47,6 +47,7 @@ def build_harness(create_latency: float = 0.0, ready_latency: float = 0.0) -> Ha
         node_wait_interval=0.01,
     ).add_all_controllers(
         lifecycle_workers=256,
+        termination_workers=128,
         termination_requeue=0.02,
         drain_requeue=0.02,
         instance_poll=0.02,

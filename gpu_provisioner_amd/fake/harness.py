@@ -86,6 +86,7 @@ class Harness:
         self,
         *,
         lifecycle_workers: int = 64,
+        termination_workers: int = 32,
         termination_requeue: float = 0.05,
         drain_requeue: float = 0.05,
         instance_poll: float = 0.05,
@@ -116,7 +117,7 @@ class Harness:
         self.termination = TerminationController(
             self.kube, self.cloud, self.recorder, self.nodes, self.nodeclaims,
             self.pods, self.volumeattachments, self.eviction_queue,
-            workers=32, drain_requeue=drain_requeue, instance_poll=instance_poll,
+            workers=termination_workers, drain_requeue=drain_requeue, instance_poll=instance_poll,
         )
         self.instance_gc = InstanceGCController(
             self.kube, self.cloud, self.recorder,
