@@ -26,7 +26,9 @@ log = logging.getLogger(__name__)
 
 ARM_SCOPE = "https://management.azure.com/.default"
 IMDS_TOKEN_URL = "http://169.254.169.254/metadata/identity/oauth2/token"
-CLIENT_ASSERTION_TYPE = "urn:ietf:params:oauth:grant-type:jwt-bearer"
+# RFC 7523 §2.2: the CLIENT-ASSERTION-TYPE urn (NOT the jwt-bearer GRANT
+# urn, which AAD rejects with AADSTS50027 in this parameter)
+CLIENT_ASSERTION_TYPE = "urn:ietf:params:oauth:client-assertion-type:jwt-bearer"
 TOKEN_FILE_CACHE_SECONDS = 300.0  # reference cred.go:125-135
 TOKEN_REFRESH_SKEW = 120.0
 

@@ -81,6 +81,12 @@ def test_workload_identity_token_exchange(tmp_path):
         body = dict(p.split("=", 1) for p in request.content.decode().split("&"))
         assert body["client_id"] == "client"
         assert "jwt-assertion" in body["client_assertion"]
+        # RFC 7523 client-assertion-type urn, URL-encoded (AAD rejects the
+        # jwt-bearer GRANT urn here)
+        assert body["client_assertion_type"] == (
+            "urn%3Aietf%3Aparams%3Aoauth%3Aclient-assertion-type%3Ajwt-bearer"
+        )
+        assert body["grant_type"] == "client_credentials"
         assert "tenant" in str(request.url)
         return httpx.Response(200, json={"access_token": "aad-token", "expires_in": 3600})
 
