@@ -12,6 +12,7 @@ from __future__ import annotations
 import copy
 import re
 from dataclasses import dataclass, field
+from fractions import Fraction
 from datetime import datetime, timezone
 from typing import Any, Iterable, Optional
 
@@ -47,7 +48,7 @@ def parse_time(s: str) -> datetime:
 _QTY_RE = re.compile(r"^([+-]?[0-9]*\.?[0-9]+)([a-zA-Z]*)$")
 _SUFFIX = {
     "": 1,
-    "m": 10**-3,
+    "m": Fraction(1, 1000),
     "k": 10**3,
     "M": 10**6,
     "G": 10**9,
@@ -64,17 +65,25 @@ _SUFFIX = {
 
 
 class Quantity:
-    """A kubernetes resource quantity. Immutable; compares by numeric value."""
+    """A kubernetes resource quantity. Immutable; compares by numeric value.
+
+    Arithmetic is EXACT (Fraction, like upstream resource.Quantity's
+    inf.Dec): float rounding on milli-quantities would make
+    allocatable-precompute results drift from what the kubelet reports."""
 
     __slots__ = ("raw", "value")
 
-    def __init__(self, raw: "str | int | float | Quantity"):
+    def __init__(self, raw: "str | int | float | Fraction | Quantity"):
         if isinstance(raw, Quantity):
             self.raw, self.value = raw.raw, raw.value
             return
+        if isinstance(raw, Fraction):
+            self.value = raw
+            self.raw = _fmt_num(raw)
+            return
         if isinstance(raw, (int, float)):
             self.raw = str(raw)
-            self.value = float(raw)
+            self.value = Fraction(raw)
             return
         m = _QTY_RE.match(str(raw))
         if not m:
@@ -83,7 +92,7 @@ class Quantity:
         if suf not in _SUFFIX:
             raise ValueError(f"invalid quantity suffix {suf!r} in {raw!r}")
         self.raw = str(raw)
-        self.value = float(num) * _SUFFIX[suf]
+        self.value = Fraction(num) * _SUFFIX[suf]
 
     def __repr__(self) -> str:
         return f"Quantity({self.raw!r})"
@@ -104,19 +113,22 @@ class Quantity:
         return hash(self.value)
 
     def __add__(self, other: "Quantity") -> "Quantity":
-        return Quantity(_fmt_num(self.value + other.value))
+        return Quantity(self.value + other.value)
 
     def __sub__(self, other: "Quantity") -> "Quantity":
-        return Quantity(_fmt_num(self.value - other.value))
+        return Quantity(self.value - other.value)
 
     def is_zero(self) -> bool:
         return self.value == 0
 
 
-def _fmt_num(v: float) -> str:
-    if v == int(v):
-        return str(int(v))
-    return str(v)
+def _fmt_num(v: Fraction) -> str:
+    if v.denominator == 1:
+        return str(v.numerator)
+    milli = v * 1000
+    if milli.denominator == 1:
+        return f"{milli.numerator}m"  # sub-unit results render as milli
+    return str(float(v))
 
 
 def qty(v: "str | int | float | Quantity") -> Quantity:

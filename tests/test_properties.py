@@ -1,0 +1,187 @@
+"""Property-based tests (hypothesis) for the algebraic cores the controllers
+lean on: requirement intersection must behave as set intersection, quantity
+parsing must round-trip and order consistently, JSON merge patch must obey
+RFC 7386, taint merge must be idempotent, and timestamps must round-trip.
+These are the pieces where a subtle algebra bug silently mis-schedules or
+mis-syncs — the reference trusts upstream karpenter's battle-testing here;
+we prove ours instead."""
+from datetime import datetime, timezone
+
+from hypothesis import given, settings, strategies as st
+
+from gpu_provisioner_amd.kube import objects as ko
+from gpu_provisioner_amd.kube.client import json_merge_patch
+from gpu_provisioner_amd.scheduling.requirements import (
+    DOES_NOT_EXIST,
+    EXISTS,
+    GT,
+    IN,
+    LT,
+    NOT_IN,
+    Requirement,
+)
+
+# -- requirement algebra ------------------------------------------------------
+
+values = st.sampled_from(["1", "2", "3", "10", "25", "a", "b", "c", "mi355x"])
+value_sets = st.lists(values, min_size=0, max_size=4)
+
+
+@st.composite
+def requirements(draw):
+    op = draw(st.sampled_from([IN, NOT_IN, EXISTS, DOES_NOT_EXIST, GT, LT]))
+    if op in (GT, LT):
+        return Requirement("k", op, [str(draw(st.integers(0, 30)))])
+    if op in (EXISTS, DOES_NOT_EXIST):
+        return Requirement("k", op)
+    vals = draw(value_sets)
+    return Requirement("k", op, vals)
+
+
+@given(a=requirements(), b=requirements(), v=values)
+@settings(max_examples=300, deadline=None)
+def test_intersection_is_set_intersection(a, b, v):
+    """(a ∩ b).has(v) == a.has(v) and b.has(v) — the defining property."""
+    assert a.intersect(b).has(v) == (a.has(v) and b.has(v))
+
+
+@given(a=requirements(), b=requirements(), v=values)
+@settings(max_examples=200, deadline=None)
+def test_intersection_commutes(a, b, v):
+    assert a.intersect(b).has(v) == b.intersect(a).has(v)
+
+
+@given(a=requirements(), b=requirements(), c=requirements(), v=values)
+@settings(max_examples=200, deadline=None)
+def test_intersection_associates(a, b, c, v):
+    assert a.intersect(b).intersect(c).has(v) == a.intersect(b.intersect(c)).has(v)
+
+
+@given(a=requirements(), v=values)
+@settings(max_examples=200, deadline=None)
+def test_intersection_idempotent(a, v):
+    assert a.intersect(a).has(v) == a.has(v)
+
+
+@given(a=requirements())
+@settings(max_examples=200, deadline=None)
+def test_requirement_dict_round_trip(a):
+    """to_dict → from_dict preserves membership semantics."""
+    b = Requirement.from_dict(a.to_dict())
+    for v in ("1", "2", "10", "25", "a", "mi355x", "zz"):
+        assert a.has(v) == b.has(v), (a.to_dict(), v)
+
+
+# -- quantities ---------------------------------------------------------------
+
+suffixes = st.sampled_from(["", "m", "k", "Ki", "M", "Mi", "G", "Gi", "T", "Ti"])
+
+
+@given(n=st.integers(0, 10**6), suf=suffixes)
+@settings(max_examples=300, deadline=None)
+def test_quantity_parse_round_trip(n, suf):
+    q = ko.qty(f"{n}{suf}")
+    assert ko.qty(str(q)) == q
+
+
+@given(
+    a=st.integers(0, 10**6), sa=suffixes,
+    b=st.integers(0, 10**6), sb=suffixes,
+)
+@settings(max_examples=300, deadline=None)
+def test_quantity_ordering_matches_values(a, sa, b, sb):
+    qa, qb = ko.qty(f"{a}{sa}"), ko.qty(f"{b}{sb}")
+    assert (qa < qb) == (qa.value < qb.value)
+    assert (qa == qb) == (qa.value == qb.value)
+
+
+@given(a=st.integers(0, 10**6), b=st.integers(0, 10**6), suf=suffixes)
+@settings(max_examples=200, deadline=None)
+def test_quantity_subtraction_consistent(a, b, suf):
+    hi, lo = max(a, b), min(a, b)
+    q = ko.qty(f"{hi}{suf}") - ko.qty(f"{lo}{suf}")
+    assert q.value == ko.qty(f"{hi - lo}{suf}").value
+
+
+# -- JSON merge patch (RFC 7386) ---------------------------------------------
+
+json_scalars = st.one_of(st.integers(-5, 5), st.text(max_size=3), st.booleans())
+json_vals = st.recursive(
+    json_scalars,
+    lambda inner: st.dictionaries(st.text(max_size=2), inner, max_size=3),
+    max_leaves=8,
+)
+json_docs = st.dictionaries(st.text(max_size=2), json_vals, max_size=4)
+# patches may include None (= delete key)
+patch_vals = st.recursive(
+    st.one_of(json_scalars, st.none()),
+    lambda inner: st.dictionaries(st.text(max_size=2), inner, max_size=3),
+    max_leaves=8,
+)
+patch_docs = st.dictionaries(st.text(max_size=2), patch_vals, max_size=4)
+
+
+def rfc7386(target, patch):
+    """Reference implementation, straight from the RFC pseudocode."""
+    if not isinstance(patch, dict):
+        return patch
+    if not isinstance(target, dict):
+        target = {}
+    out = dict(target)
+    for k, v in patch.items():
+        if v is None:
+            out.pop(k, None)
+        else:
+            out[k] = rfc7386(out.get(k), v)
+    return out
+
+
+@given(target=json_docs, patch=patch_docs)
+@settings(max_examples=300, deadline=None)
+def test_json_merge_patch_matches_rfc(target, patch):
+    assert json_merge_patch(target, patch) == rfc7386(target, patch)
+
+
+@given(target=json_docs, patch=patch_docs)
+@settings(max_examples=200, deadline=None)
+def test_json_merge_patch_does_not_mutate_inputs(target, patch):
+    import copy
+
+    t0, p0 = copy.deepcopy(target), copy.deepcopy(patch)
+    json_merge_patch(target, patch)
+    assert target == t0 and patch == p0
+
+
+# -- taints -------------------------------------------------------------------
+
+taints = st.lists(
+    st.builds(
+        lambda k, e: {"key": k, "effect": e},
+        st.sampled_from(["a", "b", "c"]),
+        st.sampled_from(["NoSchedule", "NoExecute"]),
+    ),
+    max_size=4,
+)
+
+
+@given(existing=taints, desired=taints)
+@settings(max_examples=200, deadline=None)
+def test_merge_taints_idempotent_and_complete(existing, desired):
+    once = ko.merge_taints(existing, desired)
+    twice = ko.merge_taints(once, desired)
+    assert once == twice  # idempotent
+    have = {(t["key"], t["effect"]) for t in once}
+    for t in existing + desired:
+        assert (t["key"], t["effect"]) in have
+    # existing entries keep their position and identity
+    assert once[: len(existing)] == existing
+
+
+# -- timestamps ---------------------------------------------------------------
+
+
+@given(ts=st.integers(0, 4_102_444_800))  # 1970..2100
+@settings(max_examples=300, deadline=None)
+def test_fmt_parse_time_round_trip(ts):
+    t = datetime.fromtimestamp(ts, tz=timezone.utc)
+    assert ko.parse_time(ko.fmt_time(t)) == t
