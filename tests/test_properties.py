@@ -185,3 +185,84 @@ def test_merge_taints_idempotent_and_complete(existing, desired):
 def test_fmt_parse_time_round_trip(ts):
     t = datetime.fromtimestamp(ts, tz=timezone.utc)
     assert ko.parse_time(ko.fmt_time(t)) == t
+
+
+# -- providerID parsing -------------------------------------------------------
+
+pool_names = st.from_regex(r"[a-z][a-z0-9]{0,11}", fullmatch=True)
+
+
+@given(
+    pool=pool_names,
+    sub=st.sampled_from(["sub1", "0000-1111"]),
+    rg=st.sampled_from(["rg", "MC_rg_cluster_loc"]),
+    h=st.from_regex(r"[0-9a-f]{8}", fullmatch=True),
+)
+@settings(max_examples=200, deadline=None)
+def test_provider_id_build_parse_round_trip(pool, sub, rg, h):
+    from gpu_provisioner_amd.utils.utils import (
+        build_provider_id,
+        parse_agent_pool_name_from_id,
+    )
+
+    assert parse_agent_pool_name_from_id(build_provider_id(sub, rg, pool, h)) == pool
+
+
+@given(junk=st.text(max_size=60))
+@settings(max_examples=200, deadline=None)
+def test_provider_id_parse_never_raises(junk):
+    from gpu_provisioner_amd.utils.utils import parse_agent_pool_name_from_id
+
+    parse_agent_pool_name_from_id(junk)  # None or a name, never an exception
+
+
+# -- label selectors ----------------------------------------------------------
+
+label_keys = st.sampled_from(["app", "env", "tier", "kaito.sh/workspace"])
+label_vals = st.sampled_from(["a", "b", "prod", "dev", ""])
+label_maps = st.dictionaries(label_keys, label_vals, max_size=3)
+
+
+@given(labels=label_maps, key=label_keys, val=label_vals)
+@settings(max_examples=300, deadline=None)
+def test_label_selector_semantics(labels, key, val):
+    from gpu_provisioner_amd.kube.client import LabelSelector
+
+    assert LabelSelector.parse(f"{key}={val}").matches(labels) == (
+        labels.get(key) == val
+    )
+    assert LabelSelector.parse(f"{key}!={val}").matches(labels) == (
+        labels.get(key) != val
+    )
+    assert LabelSelector.parse(key).matches(labels) == (key in labels)
+    assert LabelSelector.parse(f"!{key}").matches(labels) == (key not in labels)
+    assert LabelSelector.parse(f"{key} in (a,prod)").matches(labels) == (
+        labels.get(key) in ("a", "prod")
+    )
+
+
+@given(labels=label_maps, k1=label_keys, k2=label_keys, v=label_vals)
+@settings(max_examples=200, deadline=None)
+def test_label_selector_conjunction(labels, k1, k2, v):
+    from gpu_provisioner_amd.kube.client import LabelSelector
+
+    sel = LabelSelector.parse(f"{k1}={v},{k2}")
+    assert sel.matches(labels) == (labels.get(k1) == v and k2 in labels)
+
+
+# -- duration parsing ---------------------------------------------------------
+
+
+@given(
+    h=st.integers(0, 10), m=st.integers(0, 120), s=st.integers(0, 3600),
+)
+@settings(max_examples=200, deadline=None)
+def test_parse_duration_compositional(h, m, s):
+    from gpu_provisioner_amd.controllers.termination.controller import parse_duration
+
+    total = h * 3600 + m * 60 + s
+    got = parse_duration(f"{h}h{m}m{s}s")
+    if total == 0:
+        assert got is None  # k8s semantics: zero duration = unset here
+    else:
+        assert got is not None and got.total_seconds() == total
