@@ -31,6 +31,16 @@ def fmt_time(t: datetime) -> str:
     return f"{t.year:04d}-{t.month:02d}-{t.day:02d}T{t.hour:02d}:{t.minute:02d}:{t.second:02d}Z"
 
 
+def fmt_micro_time(t: datetime) -> str:
+    """metav1.MicroTime — Leases carry renew times at microsecond precision;
+    whole-second truncation makes short leases look expired to followers."""
+    t = t.astimezone(timezone.utc)
+    return (
+        f"{t.year:04d}-{t.month:02d}-{t.day:02d}T"
+        f"{t.hour:02d}:{t.minute:02d}:{t.second:02d}.{t.microsecond:06d}Z"
+    )
+
+
 def parse_time(s: str) -> datetime:
     # Tolerate fractional seconds and explicit offsets.
     s = s.strip()

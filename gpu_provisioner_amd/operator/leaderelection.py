@@ -64,14 +64,16 @@ class LeaderElector:
                 await on_stopped_leading()
 
     def _lease(self) -> dict:
-        now = ko.fmt_time(ko.now())
+        now = ko.fmt_micro_time(ko.now())
         return {
             "apiVersion": "coordination.k8s.io/v1",
             "kind": "Lease",
             "metadata": {"name": self.name, "namespace": self.namespace},
             "spec": {
                 "holderIdentity": self.identity,
-                "leaseDurationSeconds": int(self.lease_duration),
+                # the API field is integer seconds; a sub-second duration
+                # must not truncate to 0 (= every lease instantly expired)
+                "leaseDurationSeconds": max(1, int(self.lease_duration)),
                 "acquireTime": now,
                 "renewTime": now,
             },
@@ -112,7 +114,7 @@ class LeaderElector:
             return False
         if cur.get("spec", {}).get("holderIdentity") != self.identity:
             return False
-        cur["spec"]["renewTime"] = ko.fmt_time(ko.now())
+        cur["spec"]["renewTime"] = ko.fmt_micro_time(ko.now())
         try:
             await self.kube.update(cur)
             return True

@@ -37,15 +37,20 @@ class Manager:
         informers: Optional[InformerFactory] = None,
         required_crds: tuple = (),
         version: str = "0.1.0",
+        lease_duration: Optional[float] = None,
+        renew_interval: Optional[float] = None,
     ):
         self.client = client
         self.options = options
         self.informers = informers or InformerFactory(client)
         self.required_crds = required_crds  # [(api_version, kind)] probed via list
         self.version = version
+        self.lease_duration = lease_duration
+        self.renew_interval = renew_interval
         self.controllers: list = []
         self._elector: Optional[LeaderElector] = None
         self._server_tasks: list = []
+        self._elector_task: Optional[asyncio.Task] = None
         self._started = asyncio.Event()
 
     def register(self, *controllers) -> "Manager":
@@ -123,12 +128,18 @@ class Manager:
         self.informers.start_all()
         await self.informers.wait_for_sync()
         if self.options.leader_elect:
+            kwargs = {}
+            if self.lease_duration is not None:
+                kwargs["lease_duration"] = self.lease_duration
+            if self.renew_interval is not None:
+                kwargs["renew_interval"] = self.renew_interval
             self._elector = LeaderElector(
                 self.client,
                 self.options.leader_election_name,
                 self.options.leader_election_namespace,
+                **kwargs,
             )
-            asyncio.create_task(
+            self._elector_task = asyncio.create_task(
                 self._elector.run(self._start_controllers, self._stop_controllers),
                 name="leader-elector",
             )
@@ -150,6 +161,13 @@ class Manager:
         await asyncio.Event().wait()
 
     async def stop(self) -> None:
+        if self._elector_task is not None:
+            self._elector_task.cancel()
+            try:
+                await self._elector_task
+            except (asyncio.CancelledError, Exception):
+                pass
+            self._elector_task = None
         await self._stop_controllers()
         await self.informers.stop_all()
         for t in self._server_tasks:

@@ -1,0 +1,70 @@
+"""HA failover e2e: two full manager replicas contend for the Lease against
+one apiserver. Only the leader's controllers run (no split brain); when the
+leader dies without releasing the lease, the follower takes over after
+expiry and the fleet keeps converging."""
+import asyncio
+
+from gpu_provisioner_amd.apis import v1 as karpv1
+from gpu_provisioner_amd.fake.harness import Harness
+from gpu_provisioner_amd.main import build_manager
+from gpu_provisioner_amd.operator.options import Options
+from tests.conftest import run
+
+
+def leader_options() -> Options:
+    return Options.from_env_and_args(
+        ["--leader-elect", "--leader-election-namespace", "kube-system"], {}
+    )
+
+
+def test_two_replicas_single_leader_and_failover():
+    async def main():
+        h = Harness(node_wait_interval=0.01)  # apiserver + AKS sim only
+        m1 = build_manager(
+            h.kube, leader_options(), h.cloud.inner, version="m1"
+        )
+        m2 = build_manager(
+            h.kube, leader_options(), h.cloud.inner, version="m2"
+        )
+        for m in (m1, m2):
+            m.lease_duration, m.renew_interval = 2.0, 0.3
+
+        async def wait_for(pred, timeout=15.0):
+            deadline = asyncio.get_event_loop().time() + timeout
+            while not pred():
+                if asyncio.get_event_loop().time() > deadline:
+                    raise TimeoutError
+                await asyncio.sleep(0.02)
+
+        await m1.start(serve_http=False)
+        await wait_for(lambda: m1._elector.is_leader)
+        await m2.start(serve_http=False)
+        await asyncio.sleep(1.5)  # several renew periods
+        try:
+            # exactly one leader; the follower's controllers never started
+            assert m1._elector.is_leader and not m2._elector.is_leader
+            assert all(c.controller._tasks for c in m1.controllers)
+            assert all(not c.controller._tasks for c in m2.controllers)
+
+            nc = h.make_nodeclaim("ha1")
+            await h.kube.create(nc)
+            got = await h.wait_initialized("ha1", timeout=20)
+            assert karpv1.is_initialized(got)
+
+            # leader dies abruptly (no graceful lease release)
+            await m1.stop()
+            await wait_for(lambda: m2._elector.is_leader, timeout=20)
+            assert all(c.controller._tasks for c in m2.controllers)
+
+            # the new leader drives both new work and teardown of old work
+            await h.kube.create(h.make_nodeclaim("ha2"))
+            got2 = await h.wait_initialized("ha2", timeout=20)
+            assert karpv1.is_initialized(got2)
+            await h.kube.delete(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "ha1")
+            await h.wait_gone(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "ha1", timeout=20)
+            assert "ha1" not in h.agent_pools.pools
+            assert "ha2" in h.agent_pools.pools
+        finally:
+            await m2.stop()
+
+    run(main(), timeout=120)
