@@ -44,6 +44,7 @@ NODE_IMAGE_FAMILY_ANNOTATION_KEY = "kaito.sh/node-image-family"
 
 # well-known kube keys
 INSTANCE_TYPE_LABEL_KEY = "node.kubernetes.io/instance-type"
+ZONE_LABEL_KEY = "topology.kubernetes.io/zone"
 ARCH_LABEL_KEY = "kubernetes.io/arch"
 OS_LABEL_KEY = "kubernetes.io/os"
 HOSTNAME_LABEL_KEY = "kubernetes.io/hostname"

@@ -153,16 +153,43 @@ class InstanceProvider:
                 f"{karpv1.INSTANCE_TYPE_LABEL_KEY} requirement",
                 condition_reason="NoInstanceType",
             )
-        # prefer the cheapest catalog SKU among the allowed values; fall back
-        # to the first value for SKUs outside the catalog
-        known = [(v, self.catalog.get(v)) for v in values]
-        priced = [
-            (it.cheapest_offering().price if it.cheapest_offering() else float("inf"), v)
-            for v, it in known
-            if it is not None
-        ]
+        # karpenter semantics: among the allowed SKUs, pick the one with the
+        # cheapest offering COMPATIBLE with the claim's zone / capacity-type
+        # requirements (not just the globally cheapest); SKUs outside the
+        # catalog fall back to the first allowed value
+        from ...scheduling.requirements import Requirements
+
+        reqs = Requirements.from_nodeclaim(nodeclaim)
+        zone_req = reqs.get(karpv1.ZONE_LABEL_KEY)
+        ct_req = reqs.get(karpv1.CAPACITY_TYPE_LABEL_KEY)
+
+        def eligible_price(it) -> Optional[float]:
+            prices = [
+                o.price
+                for o in it.offerings
+                if o.available
+                and (zone_req is None or zone_req.has(o.zone))
+                and (ct_req is None or ct_req.has(o.capacity_type))
+            ]
+            return min(prices) if prices else None
+
+        priced = []
+        for v in values:
+            it = self.catalog.get(v)
+            if it is None:
+                continue
+            p = eligible_price(it)
+            if p is not None:
+                priced.append((p, v))
         if priced:
             return min(priced)[1]
+        if any(self.catalog.get(v) is not None for v in values):
+            # SKUs known but no offering satisfies the zone/capacity-type
+            # requirements — surface as capacity exhaustion so the claim is
+            # released for the owner to retry (launch error taxonomy)
+            raise InsufficientCapacityError(
+                f"no offering of {values} satisfies the zone/capacity-type requirements"
+            )
         return values[0]
 
     def new_agent_pool_object(self, nodeclaim: dict, vm_size: str) -> dict:

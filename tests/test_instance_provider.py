@@ -274,3 +274,41 @@ def test_instance_conversion_detects_deleting_state():
     # allocatable precompute from the catalog
     assert nc["status"]["capacity"][karpv1.AMD_GPU_RESOURCE] == "8"
     assert ko.qty(nc["status"]["allocatable"]["cpu"]) < ko.qty(nc["status"]["capacity"]["cpu"])
+
+
+def test_vm_size_respects_zone_and_capacity_type_requirements():
+    """Offering eligibility: the cheapest SKU overall loses to one whose
+    offerings actually satisfy the claim's zone/capacity-type requirements."""
+    from gpu_provisioner_amd.cloudprovider.types import InsufficientCapacityError
+
+    provider, _, _, _ = make_provider()
+    nc = {
+        "apiVersion": karpv1.API_VERSION,
+        "kind": karpv1.KIND_NODECLAIM,
+        "metadata": {"name": "zreq"},
+        "spec": {
+            "requirements": [
+                {
+                    "key": karpv1.INSTANCE_TYPE_LABEL_KEY,
+                    "operator": "In",
+                    "values": ["Standard_ND32is_MI355X_v6", "Standard_ND128isr_MI355X_v6"],
+                },
+                {"key": karpv1.ZONE_LABEL_KEY, "operator": "In", "values": ["eastus2-2"]},
+            ]
+        },
+    }
+    # both SKUs offer eastus2-2: cheapest (ND32) wins
+    assert provider._pick_vm_size(nc) == "Standard_ND32is_MI355X_v6"
+
+    # restrict to a zone no offering has: InsufficientCapacity, not a pick
+    nc["spec"]["requirements"][1]["values"] = ["nowhere-9"]
+    with pytest.raises(InsufficientCapacityError):
+        provider._pick_vm_size(nc)
+
+    # spot-only requirement still eligible (catalog offers spot everywhere)
+    nc["spec"]["requirements"][1] = {
+        "key": karpv1.CAPACITY_TYPE_LABEL_KEY,
+        "operator": "In",
+        "values": [karpv1.CAPACITY_TYPE_SPOT],
+    }
+    assert provider._pick_vm_size(nc) == "Standard_ND32is_MI355X_v6"
