@@ -218,6 +218,15 @@ class InstanceProvider:
         if capacity_type == karpv1.CAPACITY_TYPE_SPOT:
             props["scaleSetPriority"] = "Spot"
             props["scaleSetEvictionPolicy"] = "Delete"
+        # zone requirement → AKS availabilityZones (zone values are
+        # "<region>-<n>"; the pool API takes the bare zone numbers)
+        zones = [
+            z.rsplit("-", 1)[-1]
+            for z in karpv1.requirement_values(nodeclaim, karpv1.ZONE_LABEL_KEY)
+            if z.rsplit("-", 1)[-1].isdigit()
+        ]
+        if zones:
+            props["availabilityZones"] = sorted(set(zones))
         disk = (
             nodeclaim.get("spec", {})
             .get("resources", {})

@@ -312,3 +312,16 @@ def test_vm_size_respects_zone_and_capacity_type_requirements():
         "values": [karpv1.CAPACITY_TYPE_SPOT],
     }
     assert provider._pick_vm_size(nc) == "Standard_ND32is_MI355X_v6"
+
+
+def test_zone_requirement_sets_availability_zones():
+    provider, _, _, _ = make_provider()
+    nc = nodeclaim("zone1")
+    nc["spec"]["requirements"].append(
+        {"key": karpv1.ZONE_LABEL_KEY, "operator": "In", "values": ["eastus2-1", "eastus2-3"]}
+    )
+    pool = provider.new_agent_pool_object(nc, VM)
+    assert pool["properties"]["availabilityZones"] == ["1", "3"]
+    # no zone requirement -> field absent (AKS default zone spread)
+    pool2 = provider.new_agent_pool_object(nodeclaim("zone2"), VM)
+    assert "availabilityZones" not in pool2["properties"]
