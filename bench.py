@@ -44,7 +44,9 @@ def build_harness(create_latency: float = 0.0, ready_latency: float = 0.0) -> Ha
     return Harness(
         create_latency=create_latency,
         ready_latency=ready_latency,
-        node_wait_interval=0.01,
+        # node-wait polls hit the informer cache (zero-copy dict lookups),
+        # so a tight interval is cheap; 10ms quantized p50 Ready visibly
+        node_wait_interval=0.002,
     ).add_all_controllers(
         lifecycle_workers=256,
         termination_workers=128,

@@ -74,6 +74,12 @@ class Informer:
             except Exception:
                 log.exception("informer %s handler failed during replay", self.kind)
 
+    def remove_handler(self, fn: Callable) -> None:
+        try:
+            self._handlers.remove(fn)
+        except ValueError:
+            pass
+
     # -- cache access -------------------------------------------------------
     #
     # Shared-object contract (client-go parity): objects returned from the

@@ -243,8 +243,38 @@ class InstanceProvider:
         return {"name": ko.name_of(nodeclaim), "properties": props}
 
     async def _wait_for_node(self, pool: str) -> tuple:
-        """Poll for the Node object + providerID (reference instance.go:123-149,
-        getNodesByName :371-385 — lookup via the two agentpool labels)."""
+        """Wait for the Node object + providerID (reference instance.go:123-149,
+        getNodesByName :371-385 — lookup via the two agentpool labels).
+
+        With a Node informer the wait is EVENT-driven: a temporary handler
+        resolves as soon as a matching node (with providerID) lands in the
+        cache — add_handler's cache replay covers nodes that already exist.
+        Without one, fall back to the reference's poll loop."""
+        inf = self.nodes_informer
+        if inf is not None and inf.has_synced:
+            fut: asyncio.Future = asyncio.get_running_loop().create_future()
+
+            def on_event(event_type: str, obj: dict) -> None:
+                if fut.done() or event_type == "DELETED":
+                    return
+                labels = ko.labels_of(obj)
+                if pool in (
+                    labels.get(karpv1.AGENTPOOL_LABEL_KEY),
+                    labels.get(karpv1.AZURE_AGENTPOOL_LABEL_KEY),
+                ):
+                    pid = ko.provider_id_of(obj)
+                    if pid:
+                        fut.set_result((pid, obj))
+
+            inf.add_handler(on_event)
+            try:
+                return await asyncio.wait_for(
+                    fut, timeout=self.node_wait_attempts * self.node_wait_interval
+                )
+            except asyncio.TimeoutError:
+                return "", None
+            finally:
+                inf.remove_handler(on_event)
         for _ in range(self.node_wait_attempts):
             node = await self._node_for_pool(pool)
             if node is not None:
