@@ -142,6 +142,12 @@ class LifecycleController:
             res = await sub(nodeclaim)
             if res is not None:
                 results.append(res)
+        # a rate-limited requeue (requeue=True, ~5ms backoff) outranks any
+        # timed requeue. Dropping it loses the retry entirely and wedges the
+        # claim until an unrelated event arrives — found by the chaos suite
+        # (registration's node-patch conflict retry was being discarded).
+        if any(r.requeue for r in results):
+            return Result(requeue=True)
         requeues = [r.requeue_after for r in results if r.requeue_after is not None]
         if requeues:
             return Result(requeue_after=min(requeues))
@@ -306,8 +312,10 @@ class LifecycleController:
             await self.kube.patch("v1", "Node", ko.name_of(node), patch)
         except ConflictError:
             # node changed under us (e.g. kubelet stripped a startup taint):
-            # re-run registration against the fresh node
-            return Result(requeue_after=REGISTRATION_REQUEUE)
+            # retry on the FAST per-item backoff (5ms base) — the 1s
+            # registration requeue here put a visible tail on herd
+            # provisioning (conflicts with kubelet writes are common at scale)
+            return Result(requeue=True)
         status = nodeclaim.setdefault("status", {})
         status["nodeName"] = ko.name_of(node)
         ko.set_condition(nodeclaim, karpv1.COND_REGISTERED, ko.CONDITION_TRUE, "Registered")

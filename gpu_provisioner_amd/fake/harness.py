@@ -221,33 +221,68 @@ class Harness:
     async def wait_initialized(
         self, name: str, timeout: float = 10.0, interval: float = 0.01
     ) -> dict:
-        """Wait through the watch surface (informer cache — the same view a
-        real client like KAITO observes), falling back to apiserver reads
-        before sync."""
+        """Wait through the watch surface (the same view a real client like
+        KAITO observes): event-driven on the NodeClaim informer (handler
+        replay covers already-initialized claims), polling fallback before
+        sync. A polling wait at N-hundred concurrent waiters floods the
+        loop with timers."""
+        inf = self._informer_for(karpv1.API_VERSION, karpv1.KIND_NODECLAIM)
+        if inf is not None:
+            fut: asyncio.Future = asyncio.get_running_loop().create_future()
+
+            def on_event(event_type: str, obj: dict) -> None:
+                if (
+                    not fut.done()
+                    and event_type != "DELETED"
+                    and ko.name_of(obj) == name
+                    and karpv1.is_initialized(obj)
+                ):
+                    fut.set_result(obj)
+
+            inf.add_handler(on_event)
+            try:
+                found = await asyncio.wait_for(fut, timeout)
+            except asyncio.TimeoutError:
+                raise TimeoutError(f"NodeClaim {name} not Initialized within {timeout}s")
+            finally:
+                inf.remove_handler(on_event)
+            # private copy: callers may mutate (cache objects are shared)
+            return ko.deep_copy(found)
 
         async def check():
-            inf = self._informer_for(karpv1.API_VERSION, karpv1.KIND_NODECLAIM)
-            if inf is not None:
-                nc = inf.get(name)
-            else:
-                try:
-                    nc = await self.kube.get(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, name)
-                except Exception:
-                    return None
+            try:
+                nc = await self.kube.get(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, name)
+            except Exception:
+                return None
             return nc if nc and karpv1.is_initialized(nc) else None
 
-        found = await self.wait_for(check, timeout, interval)
-        # private copy: callers may mutate the result (cache objects are shared)
-        return ko.deep_copy(found)
+        return ko.deep_copy(await self.wait_for(check, timeout, interval))
 
     async def wait_gone(
         self, api_version: str, kind: str, name: str, timeout: float = 10.0,
         interval: float = 0.01,
     ):
+        inf = self._informer_for(api_version, kind)
+        if inf is not None:
+            if inf.get(name) is None:
+                return True
+            fut: asyncio.Future = asyncio.get_running_loop().create_future()
+
+            def on_event(event_type: str, obj: dict) -> None:
+                if not fut.done() and event_type == "DELETED" and ko.name_of(obj) == name:
+                    fut.set_result(True)
+
+            inf.add_handler(on_event)
+            try:
+                if inf.get(name) is None:  # deleted between check and park
+                    return True
+                return await asyncio.wait_for(fut, timeout)
+            except asyncio.TimeoutError:
+                raise TimeoutError(f"{kind} {name} still present after {timeout}s")
+            finally:
+                inf.remove_handler(on_event)
+
         async def check():
-            inf = self._informer_for(api_version, kind)
-            if inf is not None:
-                return True if inf.get(name) is None else None
             try:
                 await self.kube.get(api_version, kind, name)
                 return None

@@ -80,9 +80,7 @@ class Controller:
 
     def enqueue_nowait(self, key: str) -> None:
         """Synchronous enqueue for informer handlers (not coroutines)."""
-        task = asyncio.get_event_loop().create_task(self.queue.add(key))
-        # keep a reference so the task isn't GC'd before running
-        self._fire_and_forget(task)
+        self.queue.add_nowait(key)
 
     _pending: set = set()
 

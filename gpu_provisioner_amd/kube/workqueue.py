@@ -100,6 +100,7 @@ class RateLimitingQueue:
     def _wake_one(self) -> None:
         while self._getters:
             fut = self._getters.popleft()
+            fut._parked = False  # type: ignore[attr-defined]
             if not fut.done():
                 fut.set_result(None)
                 return
@@ -107,12 +108,16 @@ class RateLimitingQueue:
     def _wake_all(self) -> None:
         while self._getters:
             fut = self._getters.popleft()
+            fut._parked = False  # type: ignore[attr-defined]
             if not fut.done():
                 fut.set_result(None)
 
     # -- core ---------------------------------------------------------------
 
-    async def add(self, item: Hashable) -> None:
+    def add_nowait(self, item: Hashable) -> None:
+        """Synchronous add — the state machine has no awaits, so informer
+        handlers enqueue directly instead of spawning a task per event
+        (task-per-event was a visible constant at 128-concurrent)."""
         if self._shutdown or item in self._dirty:
             return
         self.adds += 1
@@ -120,6 +125,9 @@ class RateLimitingQueue:
         if item not in self._processing:
             self._queue.append(item)
             self._wake_one()
+
+    async def add(self, item: Hashable) -> None:
+        self.add_nowait(item)
 
     async def add_after(self, item: Hashable, delay: float) -> None:
         if delay <= 0:
@@ -148,6 +156,7 @@ class RateLimitingQueue:
                 return None
             timeout = self._next_delay()
             fut = loop.create_future()
+            fut._parked = True  # type: ignore[attr-defined]
             self._getters.append(fut)
             handle = None
             if timeout is not None:
@@ -166,10 +175,14 @@ class RateLimitingQueue:
             finally:
                 if handle is not None:
                     handle.cancel()
-                try:
-                    self._getters.remove(fut)
-                except ValueError:
-                    pass
+                # O(1) in the common case: _wake_one/_wake_all already
+                # unparked us; only a timer/cancel exit still sits in the
+                # deque (a full-deque remove per get was O(workers) each)
+                if getattr(fut, "_parked", False):
+                    try:
+                        self._getters.remove(fut)
+                    except ValueError:
+                        pass
 
     async def done(self, item: Hashable) -> None:
         self._processing.discard(item)

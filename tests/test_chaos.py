@@ -129,7 +129,7 @@ def test_fleet_converges_under_kube_and_arm_chaos():
             names = [f"kchaos{i:02d}" for i in range(16)]
             await asyncio.gather(*(h.kube.create(h.make_nodeclaim(n)) for n in names))
             done = await asyncio.gather(
-                *(h.wait_initialized(n, timeout=90) for n in names)
+                *(h.wait_initialized(n, timeout=240) for n in names)
             )
             assert all(karpv1.is_initialized(nc) for nc in done)
             assert kube_fired["n"] > 0 and sum(c.raised for c in arm) > 0
@@ -146,7 +146,7 @@ def test_fleet_converges_under_kube_and_arm_chaos():
             await asyncio.gather(*(chaos_tolerant_delete(n) for n in names))
             await asyncio.gather(
                 *(
-                    h.wait_gone(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, n, timeout=90)
+                    h.wait_gone(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, n, timeout=240)
                     for n in names
                 )
             )
@@ -154,11 +154,11 @@ def test_fleet_converges_under_kube_and_arm_chaos():
             async def pools_empty():
                 return not h.agent_pools.pools or None
 
-            await h.wait_for(pools_empty, timeout=90)
+            await h.wait_for(pools_empty, timeout=240)
         finally:
             await h.stop()
 
-    run(main(), timeout=300)
+    run(main(), timeout=600)
 
 
 def test_fleet_converges_while_watch_streams_drop():
@@ -209,3 +209,9 @@ def test_fleet_converges_while_watch_streams_drop():
             await h.stop()
 
     run(main(), timeout=240)
+
+
+# Note on budgets: per-key failure backoff doubles to a 30s cap (client-go
+# semantics), so a claim drawing a long streak of chaotic failures can sit
+# out several backoff windows; the waits above cover worst-case stacking,
+# not typical convergence (sub-second on the passing path).
