@@ -271,51 +271,56 @@ class LifecycleController:
                 await self._patch_status(nodeclaim)
             return Result(requeue_after=REGISTRATION_REQUEUE)
 
-        # sync taints/labels/owner-ref onto the node (registration.go:117-147)
-        desired_labels = {
-            **Requirements.from_nodeclaim(nodeclaim).labels(),
-            **ko.labels_of(nodeclaim),
-            karpv1.NODE_REGISTERED_LABEL_KEY: "true",
-        }
-        node_labels = {**ko.labels_of(node), **desired_labels}
-        node_taints = ko.merge_taints(
-            ko.node_taints(node), nodeclaim.get("spec", {}).get("taints") or []
-        )
-        node_finalizers = list(ko.finalizers_of(node))
-        if karpv1.TERMINATION_FINALIZER not in node_finalizers:
-            # the node carries the termination finalizer so node deletion runs
-            # the drain pipeline (node.termination controller) before the
-            # kubelet object vanishes
-            node_finalizers.append(karpv1.TERMINATION_FINALIZER)
-        patch: dict = {
-            "metadata": {
-                "labels": node_labels,
-                "finalizers": node_finalizers,
-                # optimistic lock: the taint write below is merged from OUR
-                # read of the node — without the precondition it can race
-                # the kubelet's one-shot removal of the not-ready startup
-                # taint and resurrect it forever (found by the chaos test)
-                "resourceVersion": ko.meta(node).get("resourceVersion"),
-                "ownerReferences": [
-                    {
-                        "apiVersion": karpv1.API_VERSION,
-                        "kind": karpv1.KIND_NODECLAIM,
-                        "name": ko.name_of(nodeclaim),
-                        "uid": ko.uid_of(nodeclaim),
-                        "blockOwnerDeletion": True,
-                    }
-                ],
-            },
-            "spec": {"taints": node_taints or None},
-        }
-        try:
-            await self.kube.patch("v1", "Node", ko.name_of(node), patch)
-        except ConflictError:
-            # node changed under us (e.g. kubelet stripped a startup taint):
-            # retry on the FAST per-item backoff (5ms base) — the 1s
-            # registration requeue here put a visible tail on herd
-            # provisioning (conflicts with kubelet writes are common at scale)
-            return Result(requeue=True)
+        # sync taints/labels/owner-ref onto the node (registration.go:117-147).
+        # Optimistic lock on the taint merge: without the rv precondition the
+        # write races the kubelet's one-shot removal of the not-ready startup
+        # taint and resurrects it forever (found by the chaos test). The
+        # first attempt uses the informer's node; a conflict means our cached
+        # rv is stale (the kubelet just wrote), so retry against a FRESH
+        # apiserver read instead of burning a requeue cycle per conflict —
+        # informer lag made conflict loops the dominant herd-provision tail.
+        for attempt in range(5):
+            desired_labels = {
+                **Requirements.from_nodeclaim(nodeclaim).labels(),
+                **ko.labels_of(nodeclaim),
+                karpv1.NODE_REGISTERED_LABEL_KEY: "true",
+            }
+            node_labels = {**ko.labels_of(node), **desired_labels}
+            node_taints = ko.merge_taints(
+                ko.node_taints(node), nodeclaim.get("spec", {}).get("taints") or []
+            )
+            node_finalizers = list(ko.finalizers_of(node))
+            if karpv1.TERMINATION_FINALIZER not in node_finalizers:
+                # the node carries the termination finalizer so node deletion
+                # runs the drain pipeline before the kubelet object vanishes
+                node_finalizers.append(karpv1.TERMINATION_FINALIZER)
+            patch: dict = {
+                "metadata": {
+                    "labels": node_labels,
+                    "finalizers": node_finalizers,
+                    "resourceVersion": ko.meta(node).get("resourceVersion"),
+                    "ownerReferences": [
+                        {
+                            "apiVersion": karpv1.API_VERSION,
+                            "kind": karpv1.KIND_NODECLAIM,
+                            "name": ko.name_of(nodeclaim),
+                            "uid": ko.uid_of(nodeclaim),
+                            "blockOwnerDeletion": True,
+                        }
+                    ],
+                },
+                "spec": {"taints": node_taints or None},
+            }
+            try:
+                await self.kube.patch("v1", "Node", ko.name_of(node), patch)
+                break
+            except ConflictError:
+                try:
+                    node = await self.kube.get("v1", "Node", ko.name_of(node))
+                except NotFoundError:
+                    return Result(requeue_after=REGISTRATION_REQUEUE)
+        else:
+            return Result(requeue=True)  # persistent contention: back off
         status = nodeclaim.setdefault("status", {})
         status["nodeName"] = ko.name_of(node)
         ko.set_condition(nodeclaim, karpv1.COND_REGISTERED, ko.CONDITION_TRUE, "Registered")
