@@ -8,6 +8,7 @@ import random
 
 from gpu_provisioner_amd.apis import v1 as karpv1
 from gpu_provisioner_amd.fake.harness import Harness
+from gpu_provisioner_amd.kube import objects as ko
 from gpu_provisioner_amd.providers.instance.armapi import ARMError
 from tests.conftest import run
 
@@ -215,3 +216,103 @@ def test_fleet_converges_while_watch_streams_drop():
 # semantics), so a claim drawing a long streak of chaotic failures can sit
 # out several backoff windows; the waits above cover worst-case stacking,
 # not typical convergence (sub-second on the passing path).
+
+
+def test_full_topology_chaos_with_repair_and_drift():
+    """All controllers on (health repair with short tolerations, drift
+    replacement, both GCs) while the environment misbehaves: GPU health
+    conditions flap sick/healthy, pools mutate out-of-band (drift), and
+    claims are deleted mid-life. The fleet must end with every surviving
+    claim Initialized on a conforming pool, and teardown must leave
+    nothing behind."""
+    from gpu_provisioner_amd.nodeagent import (
+        GPUReport,
+        NodeReport,
+        patch_node_condition,
+    )
+
+    def sick():
+        return NodeReport(healthy=False, gpu_count=8,
+                          gpus=[GPUReport(index=0, problems=["MFMA fail"])],
+                          problems=["gpu0: MFMA fail"])
+
+    def healthy():
+        return NodeReport(healthy=True, gpu_count=8,
+                          gpus=[GPUReport(index=0, healthy=True)])
+
+    async def main():
+        rng = random.Random(99)
+        h = Harness(
+            node_wait_interval=0.005, gpu_repair_toleration=0.4,
+        ).add_all_controllers(
+            gc_interval=0.5, adoption_age=0.3, drift_interval=0.2,
+            drift_replace=True, termination_requeue=0.01, drain_requeue=0.01,
+            instance_poll=0.01,
+        )
+        await h.start()
+        try:
+            names = [f"topo{i:02d}" for i in range(12)]
+            await asyncio.gather(*(h.kube.create(h.make_nodeclaim(n)) for n in names))
+            await asyncio.gather(*(h.wait_initialized(n, timeout=60) for n in names))
+
+            # 2 seconds of mayhem
+            for _ in range(10):
+                await asyncio.sleep(0.2)
+                n = rng.choice(names)
+                action = rng.random()
+                try:
+                    if action < 0.4 and n in h.agent_pools.pools:
+                        # out-of-band pool mutation → drift → replacement
+                        h.agent_pools.pools[n]["properties"]["osSKU"] = "AzureLinux"
+                    elif action < 0.7:
+                        nc = await h.kube.get(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, n)
+                        node_name = nc.get("status", {}).get("nodeName")
+                        if node_name:
+                            # GPU sickness flap: sick now, healthy shortly after
+                            await patch_node_condition(h.kube, node_name, sick())
+                            if rng.random() < 0.5:
+                                await asyncio.sleep(0.1)
+                                await patch_node_condition(h.kube, node_name, healthy())
+                    else:
+                        await h.kube.delete(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, n)
+                except Exception:
+                    pass  # races with replacement are part of the chaos
+
+            # quiesce: every surviving claim must be Initialized on a
+            # conforming (non-drifted) pool; every pool must belong to a claim
+            async def settled():
+                claims = [
+                    nc for nc in await h.kube.list(karpv1.API_VERSION, karpv1.KIND_NODECLAIM)
+                    if not ko.is_deleting(nc)
+                ]
+                if any(not karpv1.is_initialized(nc) for nc in claims):
+                    return None
+                live = {ko.name_of(nc) for nc in claims}
+                if set(h.agent_pools.pools) != live:
+                    return None
+                for name in live:
+                    if h.agent_pools.pools[name]["properties"].get("osSKU") == "AzureLinux":
+                        return None  # drifted pool still awaiting replacement
+                return claims
+
+            claims = await h.wait_for(settled, timeout=120, interval=0.1)
+
+            # full teardown
+            await asyncio.gather(
+                *(
+                    h.kube.delete(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, ko.name_of(nc))
+                    for nc in claims
+                )
+            )
+
+            async def empty():
+                pools = not h.agent_pools.pools
+                claims_left = await h.kube.list(karpv1.API_VERSION, karpv1.KIND_NODECLAIM)
+                nodes_left = await h.kube.list("v1", "Node")
+                return (pools and not claims_left and not nodes_left) or None
+
+            await h.wait_for(empty, timeout=120, interval=0.1)
+        finally:
+            await h.stop()
+
+    run(main(), timeout=400)
