@@ -228,24 +228,15 @@ class Harness:
         loop with timers."""
         inf = self._informer_for(karpv1.API_VERSION, karpv1.KIND_NODECLAIM)
         if inf is not None:
-            fut: asyncio.Future = asyncio.get_running_loop().create_future()
+            def pred(event_type: str, obj):
+                if event_type in ("DELETED", "ABSENT") or obj is None:
+                    return None
+                return obj if karpv1.is_initialized(obj) else None
 
-            def on_event(event_type: str, obj: dict) -> None:
-                if (
-                    not fut.done()
-                    and event_type != "DELETED"
-                    and ko.name_of(obj) == name
-                    and karpv1.is_initialized(obj)
-                ):
-                    fut.set_result(obj)
-
-            inf.add_handler(on_event)
             try:
-                found = await asyncio.wait_for(fut, timeout)
+                found = await inf.wait_until(pred, name=name, timeout=timeout)
             except asyncio.TimeoutError:
                 raise TimeoutError(f"NodeClaim {name} not Initialized within {timeout}s")
-            finally:
-                inf.remove_handler(on_event)
             # private copy: callers may mutate (cache objects are shared)
             return ko.deep_copy(found)
 
@@ -264,23 +255,13 @@ class Harness:
     ):
         inf = self._informer_for(api_version, kind)
         if inf is not None:
-            if inf.get(name) is None:
-                return True
-            fut: asyncio.Future = asyncio.get_running_loop().create_future()
+            def pred(event_type: str, obj):
+                return True if event_type in ("DELETED", "ABSENT") else None
 
-            def on_event(event_type: str, obj: dict) -> None:
-                if not fut.done() and event_type == "DELETED" and ko.name_of(obj) == name:
-                    fut.set_result(True)
-
-            inf.add_handler(on_event)
             try:
-                if inf.get(name) is None:  # deleted between check and park
-                    return True
-                return await asyncio.wait_for(fut, timeout)
+                return await inf.wait_until(pred, name=name, timeout=timeout)
             except asyncio.TimeoutError:
                 raise TimeoutError(f"{kind} {name} still present after {timeout}s")
-            finally:
-                inf.remove_handler(on_event)
 
         async def check():
             try:

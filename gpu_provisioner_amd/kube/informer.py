@@ -45,6 +45,8 @@ class Informer:
         self._cache: dict = {}  # key -> obj
         self._indexes: dict = {}  # index_name -> (fn, {value: set(keys)})
         self._handlers: list = []  # fn(event_type, obj)
+        self._key_waiters: dict = {}  # key -> [(predicate, future)]
+        self._index_waiters: dict = {}  # index -> {value: [(predicate, future)]}
         self._synced = asyncio.Event()
         self._task: Optional[asyncio.Task] = None
         self._rv = ""
@@ -79,6 +81,83 @@ class Informer:
             self._handlers.remove(fn)
         except ValueError:
             pass
+
+    # -- keyed waiters --------------------------------------------------------
+    #
+    # A parked plain handler costs O(waiters) per event; at a few hundred
+    # concurrent waiters that dominated the profile. Keyed waiters dispatch
+    # in O(1): by object key, or by an index value (e.g. the agentpool
+    # label) when the object's name isn't known in advance.
+
+    async def wait_until(
+        self,
+        predicate: Callable,
+        *,
+        name: str = "",
+        namespace: str = "",
+        index: str = "",
+        value: str = "",
+        timeout: float = 10.0,
+    ):
+        """Await predicate(event_type, obj) returning non-None for events on
+        ONE key — an object name or an (index, value) pair. The current
+        cache state is checked first (event_type "ADDED", or "ABSENT" with
+        obj=None for a missing name)."""
+        if name:
+            key = f"{namespace}/{name}" if namespace else name
+            obj = self._cache.get(key)
+            res = predicate("ADDED" if obj is not None else "ABSENT", obj)
+            if res is not None:
+                return res
+            waiters = self._key_waiters.setdefault(key, [])
+        else:
+            _, idx = self._indexes[index]
+            for k in list(idx.get(value, ())):
+                obj = self._cache.get(k)
+                if obj is not None:
+                    res = predicate("ADDED", obj)
+                    if res is not None:
+                        return res
+            waiters = self._index_waiters.setdefault(index, {}).setdefault(value, [])
+
+        fut: asyncio.Future = asyncio.get_running_loop().create_future()
+        entry = (predicate, fut)
+        waiters.append(entry)
+        try:
+            return await asyncio.wait_for(fut, timeout)
+        finally:
+            try:
+                waiters.remove(entry)
+            except ValueError:
+                pass
+
+    def _fire_waiters(self, waiters: Optional[list], event_type: str, obj: dict) -> None:
+        if not waiters:
+            return
+        for pred, fut in list(waiters):
+            if fut.done():
+                continue
+            try:
+                res = pred(event_type, obj)
+            except Exception:
+                log.exception("informer %s waiter predicate failed", self.kind)
+                continue
+            if res is not None:
+                fut.set_result(res)
+
+    def _dispatch_waiters(self, event_type: str, obj: dict) -> None:
+        if self._key_waiters:
+            self._fire_waiters(self._key_waiters.get(object_key(obj)), event_type, obj)
+        if self._index_waiters:
+            for index, wmap in self._index_waiters.items():
+                fn, _ = self._indexes[index]
+                vals = fn(obj)
+                if vals is None:
+                    continue
+                if isinstance(vals, str):
+                    vals = [vals]
+                for v in vals:
+                    self._fire_waiters(wmap.get(v), event_type, obj)
 
     # -- cache access -------------------------------------------------------
     #
@@ -217,6 +296,7 @@ class Informer:
                 h(event_type, obj)
             except Exception:
                 log.exception("informer %s handler failed", self.kind)
+        self._dispatch_waiters(event_type, obj)
 
 
 class InformerFactory:

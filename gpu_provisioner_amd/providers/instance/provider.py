@@ -252,29 +252,22 @@ class InstanceProvider:
         Without one, fall back to the reference's poll loop."""
         inf = self.nodes_informer
         if inf is not None and inf.has_synced:
-            fut: asyncio.Future = asyncio.get_running_loop().create_future()
+            def pred(event_type: str, obj):
+                if event_type in ("DELETED", "ABSENT") or obj is None:
+                    return None
+                pid = ko.provider_id_of(obj)
+                return (pid, obj) if pid else None
 
-            def on_event(event_type: str, obj: dict) -> None:
-                if fut.done() or event_type == "DELETED":
-                    return
-                labels = ko.labels_of(obj)
-                if pool in (
-                    labels.get(karpv1.AGENTPOOL_LABEL_KEY),
-                    labels.get(karpv1.AZURE_AGENTPOOL_LABEL_KEY),
-                ):
-                    pid = ko.provider_id_of(obj)
-                    if pid:
-                        fut.set_result((pid, obj))
-
-            inf.add_handler(on_event)
             try:
-                return await asyncio.wait_for(
-                    fut, timeout=self.node_wait_attempts * self.node_wait_interval
+                # keyed on the agentpool index: O(1) dispatch per node event
+                return await inf.wait_until(
+                    pred,
+                    index="agentpool",
+                    value=pool,
+                    timeout=self.node_wait_attempts * self.node_wait_interval,
                 )
             except asyncio.TimeoutError:
                 return "", None
-            finally:
-                inf.remove_handler(on_event)
         for _ in range(self.node_wait_attempts):
             node = await self._node_for_pool(pool)
             if node is not None:
