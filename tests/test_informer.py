@@ -133,3 +133,72 @@ def test_resync_redelivers_cached_objects():
         assert len(resyncs) >= 2, events
 
     run(main())
+
+
+def test_wait_until_by_name_and_index():
+    from gpu_provisioner_amd.fake.apiserver import InMemoryAPIServer, InMemoryClient
+    from gpu_provisioner_amd.kube.informer import Informer
+
+    async def main():
+        kube = InMemoryClient(InMemoryAPIServer())
+        inf = Informer(kube, "v1", "Node")
+        inf.add_index("pool", lambda o: (o["metadata"].get("labels") or {}).get("agentpool"))
+        inf.start()
+        await inf.wait_for_sync()
+
+        # name-keyed: resolves when the predicate matches a later event
+        async def create_later():
+            await asyncio.sleep(0.02)
+            await kube.create(
+                {"apiVersion": "v1", "kind": "Node",
+                 "metadata": {"name": "w1", "labels": {"agentpool": "p1"}},
+                 "spec": {}, "status": {}}
+            )
+
+        t = asyncio.create_task(create_later())
+        got = await inf.wait_until(
+            lambda et, o: o if o is not None and et != "DELETED" else None,
+            name="w1", timeout=5,
+        )
+        assert got["metadata"]["name"] == "w1"
+        await t
+
+        # cache-satisfied immediately (no event needed)
+        got2 = await inf.wait_until(
+            lambda et, o: o if o is not None else None, name="w1", timeout=1
+        )
+        assert got2["metadata"]["name"] == "w1"
+
+        # index-keyed: a second node in the pool arrives later
+        async def create_p2():
+            await asyncio.sleep(0.02)
+            await kube.create(
+                {"apiVersion": "v1", "kind": "Node",
+                 "metadata": {"name": "w2", "labels": {"agentpool": "p2"}},
+                 "spec": {"providerID": "azure:///x"}, "status": {}}
+            )
+
+        t2 = asyncio.create_task(create_p2())
+        got3 = await inf.wait_until(
+            lambda et, o: o if o is not None and o["spec"].get("providerID") else None,
+            index="pool", value="p2", timeout=5,
+        )
+        assert got3["metadata"]["name"] == "w2"
+        await t2
+
+        # ABSENT short-circuit (wait_gone shape)
+        gone = await inf.wait_until(
+            lambda et, o: True if et in ("DELETED", "ABSENT") else None,
+            name="nope", timeout=1,
+        )
+        assert gone is True
+
+        # timeout path
+        import pytest as _pytest
+        with _pytest.raises(asyncio.TimeoutError):
+            await inf.wait_until(lambda et, o: None, name="w1", timeout=0.05)
+        # waiter registries drained after completion/timeouts
+        assert not any(inf._key_waiters.values())
+        await inf.stop()
+
+    run(main())
