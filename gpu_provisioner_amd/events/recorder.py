@@ -10,6 +10,7 @@ import asyncio
 import logging
 import time
 import uuid
+from collections import OrderedDict
 from typing import Optional
 
 from ..kube import objects as ko
@@ -25,7 +26,10 @@ class EventRecorder:
         self.client = client
         self.namespace = namespace
         self.source = source
-        self._seen: dict = {}  # dedupe key -> monotonic deadline
+        # dedupe key -> monotonic deadline, kept in deadline order (constant
+        # TTL ⇒ insertion order == expiry order) so expiry is O(1) pops from
+        # the front — a full-dict prune per publish degraded long soaks
+        self._seen: OrderedDict = OrderedDict()
         self._pending: set = set()
 
     def publish(
@@ -39,12 +43,16 @@ class EventRecorder:
     ) -> None:
         key = (ko.uid_of(obj) or ko.name_of(obj), reason, event_type, tuple(dedupe_values))
         nw = time.monotonic()
+        while self._seen:
+            _, head = next(iter(self._seen.items()))
+            if head > nw:
+                break
+            self._seen.popitem(last=False)
         deadline = self._seen.get(key)
         if deadline is not None and nw < deadline:
             return
         self._seen[key] = nw + DEDUPE_TTL
-        if len(self._seen) > 8192:
-            self._seen = {k: v for k, v in self._seen.items() if v > nw}
+        self._seen.move_to_end(key)
         task = asyncio.get_event_loop().create_task(
             self._emit(obj, reason, message, event_type)
         )

@@ -19,6 +19,7 @@ simulated device plugin) can be tested fully in-process:
 from __future__ import annotations
 
 import asyncio
+from collections import deque
 import itertools
 import uuid
 from collections import defaultdict
@@ -63,7 +64,9 @@ class InMemoryAPIServer:
         self._rv = itertools.count(1)
         self._lock = asyncio.Lock()
         self._watchers: list = []  # (gvk, queue)
-        self._history: list = []  # (rv:int, gvk, event_type, obj)
+        # bounded ring: a list re-slice per event past the cap is an
+        # O(cap) copy per write and degrades long runs
+        self._history: deque = deque(maxlen=_WATCH_HISTORY)  # (rv, gvk, event_type, obj)
         # test hooks: fn(verb, gvk, obj_or_name) -> Optional[APIError] raised if returned
         self.reactors: list = []
         # eviction hook: fn(pod) -> Optional[APIError]
@@ -89,8 +92,6 @@ class InMemoryAPIServer:
         # stored revision in place (every write replaces it).
         rv = int(obj["metadata"]["resourceVersion"])
         self._history.append((rv, gvk, event_type, obj))
-        if len(self._history) > _WATCH_HISTORY:
-            self._history = self._history[-_WATCH_HISTORY:]
         for wgvk, queue in self._watchers:
             if wgvk == gvk:
                 queue.put_nowait((event_type, obj))
