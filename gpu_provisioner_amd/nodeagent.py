@@ -30,6 +30,9 @@ EXPECTED_HBM_GB = 288
 # for box-to-box variance; xGMI 7 links x ~153 GB/s guide values)
 MIN_HBM_BW_GBS = 4800.0
 MIN_XGMI_BW_GBS = 30.0
+# D2D copy path (SDMA/blit — the hardware RCCL's transports lean on);
+# measured ~4.9 TB/s on healthy silicon (profiles/), floor leaves margin
+MIN_SDMA_BW_GBS = 3000.0
 
 
 class NodeAgentError(RuntimeError):
@@ -60,6 +63,7 @@ class GPUReport:
     mfma_fp8_ok: bool = False
     lds_ok: bool = False
     lds_bytes_tested: int = 0
+    sdma_bw_gbs: float = 0.0
     healthy: bool = False
     problems: list = field(default_factory=list)
 
@@ -130,6 +134,14 @@ class NodeAgent:
         rc = self.lib.na_lds_selftest(dev, ctypes.byref(tested))
         return rc == 0, tested.value
 
+    def sdma_bandwidth(self, dev: int, bytes_: int = 512 << 20, iters: int = 10) -> float:
+        """D2D copy through the SDMA engines — RCCL's xGMI transport hardware."""
+        out = ctypes.c_double(0)
+        rc = self.lib.na_sdma_bandwidth(dev, ctypes.c_longlong(bytes_), iters, ctypes.byref(out))
+        if rc != 0:
+            raise NodeAgentError(f"sdma_bandwidth({dev}): {self._err()}")
+        return out.value
+
     def p2p_matrix(self, n: int) -> list:
         buf = (ctypes.c_int * (n * n))()
         if self.lib.na_p2p_matrix(n, buf) != 0:
@@ -181,6 +193,11 @@ class NodeAgent:
                 g.lds_ok, g.lds_bytes_tested = self.lds_selftest(d)
                 if not g.lds_ok:
                     g.problems.append(f"LDS selftest failed: {self._err()}")
+                g.sdma_bw_gbs = round(self.sdma_bandwidth(d), 1)
+                if g.sdma_bw_gbs < MIN_SDMA_BW_GBS:
+                    g.problems.append(
+                        f"SDMA D2D bandwidth {g.sdma_bw_gbs} GB/s below floor {MIN_SDMA_BW_GBS}"
+                    )
             except NodeAgentError as e:
                 g.problems.append(str(e))
             g.healthy = not g.problems

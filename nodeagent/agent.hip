@@ -329,6 +329,44 @@ extern "C" int na_lds_selftest(int dev, long long* bytes_tested) {
     return NA_OK;
 }
 
+extern "C" int na_sdma_bandwidth(int dev, long long bytes, int iters, double* gbs) {
+    // device-to-device copy through hipMemcpyAsync: the runtime routes
+    // large D2D copies through the SDMA engines — the same hardware RCCL
+    // leans on for xGMI peer traffic. A sick SDMA engine shows up here
+    // before any collective does.
+    HIP_CHECK(hipSetDevice(dev));
+    void *src = nullptr, *dst = nullptr;
+    HIP_CHECK(hipMalloc(&src, (size_t)bytes));
+    hipError_t e2 = hipMalloc(&dst, (size_t)bytes);
+    if (e2 != hipSuccess) {
+        (void)hipFree(src);
+        std::snprintf(na_last_error_buf, sizeof(na_last_error_buf), "%s", hipGetErrorString(e2));
+        return NA_ERR_HIP;
+    }
+    HIP_CHECK(hipMemset(src, 7, (size_t)bytes));
+    hipStream_t stream;
+    HIP_CHECK(hipStreamCreate(&stream));
+    HIP_CHECK(hipMemcpyAsync(dst, src, (size_t)bytes, hipMemcpyDeviceToDevice, stream));  // warmup
+    HIP_CHECK(hipStreamSynchronize(stream));
+    hipEvent_t t0, t1;
+    HIP_CHECK(hipEventCreate(&t0));
+    HIP_CHECK(hipEventCreate(&t1));
+    HIP_CHECK(hipEventRecord(t0, stream));
+    for (int i = 0; i < iters; ++i)
+        HIP_CHECK(hipMemcpyAsync(dst, src, (size_t)bytes, hipMemcpyDeviceToDevice, stream));
+    HIP_CHECK(hipEventRecord(t1, stream));
+    HIP_CHECK(hipEventSynchronize(t1));
+    float ms = 0.f;
+    HIP_CHECK(hipEventElapsedTime(&ms, t0, t1));
+    *gbs = (2.0 * (double)bytes * iters) / (ms * 1e6);  // read + write
+    (void)hipEventDestroy(t0);
+    (void)hipEventDestroy(t1);
+    (void)hipStreamDestroy(stream);
+    (void)hipFree(src);
+    (void)hipFree(dst);
+    return NA_OK;
+}
+
 extern "C" int na_p2p_matrix(int n, int* out /* n*n */) {
     for (int i = 0; i < n; ++i) {
         for (int j = 0; j < n; ++j) {

@@ -35,6 +35,7 @@ def test_hip_library_builds_and_loads():
         "na_mfma_bf16_selftest",
         "na_mfma_fp8_selftest",
         "na_lds_selftest",
+        "na_sdma_bandwidth",
         "na_p2p_matrix",
         "na_p2p_bandwidth",
         "na_last_error",
@@ -110,6 +111,18 @@ def test_gpu_lds_selftest():
     ok, tested = agent.lds_selftest(0)
     assert ok, "LDS selftest failed"
     assert tested >= 64 * 1024, f"only {tested} LDS bytes tested"
+
+
+@pytest.mark.gpu
+def test_gpu_sdma_bandwidth():
+    """D2D copy through the SDMA engines (RCCL's xGMI transport hardware)."""
+    from gpu_provisioner_amd.nodeagent import NodeAgent
+
+    _ensure_lib()
+    agent = NodeAgent()
+    bw = agent.sdma_bandwidth(0)
+    # measured ~4.9 TB/s healthy
+    assert bw > 3000.0, f"SDMA D2D bandwidth {bw:.1f} GB/s below floor"
 
 
 @pytest.mark.gpu
