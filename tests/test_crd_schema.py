@@ -1,0 +1,107 @@
+"""Code ↔ CRD contract: objects the controllers write (and the shipped
+examples) must validate against the NodeClaim CRD's openAPIV3Schema. A
+minimal structural validator (type/properties/required/items/enum) is
+enough to catch drift between the Python surface and the CRD the chart
+installs."""
+import asyncio
+import os
+
+import pytest
+import yaml
+
+from gpu_provisioner_amd.apis import v1 as karpv1
+from gpu_provisioner_amd.fake.harness import Harness
+from tests.conftest import run
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+CRD_PATH = os.path.join(ROOT, "charts", "gpu-provisioner-amd", "crds", "karpenter.sh_nodeclaims.yaml")
+
+
+def load_schema() -> dict:
+    crd = next(yaml.safe_load_all(open(CRD_PATH)))
+    versions = crd["spec"]["versions"]
+    v1 = [v for v in versions if v["name"] == "v1"][0]
+    return v1["schema"]["openAPIV3Schema"]
+
+
+def validate(obj, schema, path="$"):
+    """Minimal openAPI v3 structural check; returns a list of violations."""
+    errs = []
+    t = schema.get("type")
+    if t == "object" or ("properties" in schema and t is None):
+        if not isinstance(obj, dict):
+            return [f"{path}: expected object, got {type(obj).__name__}"]
+        props = schema.get("properties", {})
+        extra_ok = (
+            schema.get("x-kubernetes-preserve-unknown-fields")
+            or "additionalProperties" in schema
+            or not props
+        )
+        for k, v in obj.items():
+            if k in props:
+                errs += validate(v, props[k], f"{path}.{k}")
+            elif isinstance(schema.get("additionalProperties"), dict):
+                errs += validate(v, schema["additionalProperties"], f"{path}.{k}")
+            elif not extra_ok:
+                errs.append(f"{path}.{k}: unknown field")
+        for req in schema.get("required", []):
+            if req not in obj:
+                errs.append(f"{path}.{req}: required field missing")
+    elif t == "array":
+        if not isinstance(obj, list):
+            return [f"{path}: expected array, got {type(obj).__name__}"]
+        for i, item in enumerate(obj):
+            errs += validate(item, schema.get("items", {}), f"{path}[{i}]")
+    elif t == "string":
+        if not isinstance(obj, str):
+            errs.append(f"{path}: expected string, got {type(obj).__name__}")
+        elif "enum" in schema and obj not in schema["enum"]:
+            errs.append(f"{path}: {obj!r} not in enum {schema['enum']}")
+    elif t == "integer":
+        if not isinstance(obj, int) or isinstance(obj, bool):
+            errs.append(f"{path}: expected integer, got {type(obj).__name__}")
+    elif t == "boolean":
+        if not isinstance(obj, bool):
+            errs.append(f"{path}: expected boolean, got {type(obj).__name__}")
+    # anyOf (e.g. intstr quantities): pass if any branch passes
+    if "anyOf" in schema:
+        branches = [validate(obj, b, path) for b in schema["anyOf"]]
+        if all(b for b in branches):
+            errs.append(f"{path}: no anyOf branch matched")
+    return errs
+
+
+def test_lifecycle_written_nodeclaim_matches_crd_schema():
+    schema = load_schema()
+
+    async def main():
+        h = Harness().add_all_controllers(gc_interval=60.0)
+        await h.start()
+        try:
+            await h.kube.create(h.make_nodeclaim("crd1"))
+            nc = await h.wait_initialized("crd1")
+        finally:
+            await h.stop()
+        # strip server-side metadata the schema doesn't model
+        errs = validate(nc, schema)
+        assert not errs, "\n".join(errs)
+
+    run(main())
+
+
+def test_example_nodeclaims_match_crd_schema():
+    schema = load_schema()
+    for fname in ("v1-nodeclaim-mi355x.yaml", "azure-linux-annotation-nodeclaim.yaml"):
+        path = os.path.join(ROOT, "examples", fname)
+        for doc in yaml.safe_load_all(open(path)):
+            if not doc or doc.get("kind") != "NodeClaim":
+                continue
+            errs = validate(doc, schema)
+            assert not errs, f"{fname}:\n" + "\n".join(errs)
+
+
+def test_validator_rejects_malformed():
+    schema = load_schema()
+    bad = karpv1.new_nodeclaim("bad1")
+    bad["spec"] = {"requirements": [{"key": 42, "operator": "In", "values": "notalist"}]}
+    assert validate(bad, schema), "validator failed to flag malformed requirements"
