@@ -91,7 +91,11 @@ def test_lifecycle_written_nodeclaim_matches_crd_schema():
 
 def test_example_nodeclaims_match_crd_schema():
     schema = load_schema()
-    for fname in ("v1-nodeclaim-mi355x.yaml", "azure-linux-annotation-nodeclaim.yaml"):
+    for fname in (
+        "v1-nodeclaim-mi355x.yaml",
+        "azure-linux-annotation-nodeclaim.yaml",
+        "spot-nodeclaim-mi355x.yaml",
+    ):
         path = os.path.join(ROOT, "examples", fname)
         for doc in yaml.safe_load_all(open(path)):
             if not doc or doc.get("kind") != "NodeClaim":
