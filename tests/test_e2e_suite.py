@@ -464,3 +464,45 @@ def test_metrics_wired_through_churn():
         assert sum(b.get() for b in h1._buckets) >= 1
 
     run(main())
+
+
+def test_spec14_spot_zone_pinned_claim():
+    """The spot example's shape end-to-end: spot capacity type + zone
+    pinning land on the pool (Spot priority, availabilityZones) and the
+    cheapest eligible SKU is chosen."""
+
+    async def main():
+        h = env()
+        await h.start()
+        try:
+            nc = karpv1.new_nodeclaim(
+                "spot1", labels={karpv1.KAITO_WORKSPACE_LABEL_KEY: "batch"}
+            )
+            nc["metadata"]["annotations"] = {karpv1.DO_NOT_DISRUPT_ANNOTATION_KEY: "true"}
+            nc["spec"] = {
+                "nodeClassRef": {"group": "kaito.sh", "kind": "KaitoNodeClass", "name": "default"},
+                "requirements": [
+                    {"key": karpv1.INSTANCE_TYPE_LABEL_KEY, "operator": "In",
+                     "values": ["Standard_ND64is_MI355X_v6", "Standard_ND32is_MI355X_v6"]},
+                    {"key": karpv1.CAPACITY_TYPE_LABEL_KEY, "operator": "In", "values": ["spot"]},
+                    {"key": karpv1.ZONE_LABEL_KEY, "operator": "In",
+                     "values": ["eastus2-1", "eastus2-2"]},
+                ],
+                "resources": {"requests": {karpv1.AMD_GPU_RESOURCE: "2"}},
+            }
+            await h.kube.create(nc)
+            got = await h.wait_initialized("spot1")
+            props = h.agent_pools.pools["spot1"]["properties"]
+            assert props["vmSize"] == "Standard_ND32is_MI355X_v6"  # cheapest eligible
+            assert props["scaleSetPriority"] == "Spot"
+            assert props["scaleSetEvictionPolicy"] == "Delete"
+            assert props["availabilityZones"] == ["1", "2"]
+            assert got["status"]["allocatable"][karpv1.AMD_GPU_RESOURCE] == "2"
+            labels = ko.labels_of(
+                await h.kube.get("v1", "Node", got["status"]["nodeName"])
+            )
+            assert labels[karpv1.CAPACITY_TYPE_LABEL_KEY] == "spot"
+        finally:
+            await h.stop()
+
+    run(main())
