@@ -160,6 +160,18 @@ class InstanceProvider:
         from ...scheduling.requirements import Requirements
 
         reqs = Requirements.from_nodeclaim(nodeclaim)
+        it_req = reqs.get(karpv1.INSTANCE_TYPE_LABEL_KEY)
+        if it_req is not None and it_req.min_values is not None:
+            # karpenter minValues semantics: the scheduler promised at least
+            # N instance-type options; fewer orderable ones than that means
+            # the request's flexibility contract cannot be met
+            orderable = [v for v in values if self.catalog.get(v) is not None]
+            if len(orderable) < it_req.min_values:
+                raise InsufficientCapacityError(
+                    f"instance-type requirement needs minValues="
+                    f"{it_req.min_values} orderable types, only {len(orderable)} "
+                    f"of {values} are in the catalog"
+                )
         zone_req = reqs.get(karpv1.ZONE_LABEL_KEY)
         ct_req = reqs.get(karpv1.CAPACITY_TYPE_LABEL_KEY)
 

@@ -325,3 +325,25 @@ def test_zone_requirement_sets_availability_zones():
     # no zone requirement -> field absent (AKS default zone spread)
     pool2 = provider.new_agent_pool_object(nodeclaim("zone2"), VM)
     assert "availabilityZones" not in pool2["properties"]
+
+
+def test_min_values_gates_instance_type_flexibility():
+    """karpenter minValues: the claim demands >= N orderable instance-type
+    options; fewer in the catalog -> InsufficientCapacity (claim released)."""
+    from gpu_provisioner_amd.cloudprovider.types import InsufficientCapacityError
+
+    provider, _, _, _ = make_provider()
+    nc = nodeclaim("minv1")
+    nc["spec"]["requirements"] = [
+        {
+            "key": karpv1.INSTANCE_TYPE_LABEL_KEY,
+            "operator": "In",
+            "values": [VM, "Standard_FAKE_SKU_v9"],
+            "minValues": 2,
+        }
+    ]
+    with pytest.raises(InsufficientCapacityError):
+        provider._pick_vm_size(nc)
+    # satisfied when enough orderable SKUs exist
+    nc["spec"]["requirements"][0]["values"] = [VM, "Standard_ND64is_MI355X_v6"]
+    assert provider._pick_vm_size(nc) in (VM, "Standard_ND64is_MI355X_v6")
