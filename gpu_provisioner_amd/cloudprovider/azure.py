@@ -30,6 +30,7 @@ GPU_REPAIR_TOLERATION_SECONDS = 300.0  # 5 min
 DRIFT_INSTANCE_TYPE = "InstanceTypeDrift"
 DRIFT_NODE_IMAGE = "NodeImageDrift"
 DRIFT_SKU_RETIRED = "SKURetiredDrift"
+DRIFT_REQUIREMENTS = "RequirementsDrift"
 
 
 class AzureCloudProvider(CloudProvider):
@@ -106,7 +107,25 @@ class AzureCloudProvider(CloudProvider):
             and self.catalog.is_gpu_sku(instance.type)
         ):
             return DRIFT_SKU_RETIRED
+        # RequirementsDrift: the backing Node's labels mutated out from
+        # under the claim's requirements (upstream karpenter's
+        # NodeRequirementDrift). Only keys PRESENT on the node are judged —
+        # absence may just mean the platform doesn't stamp that label.
+        node = await self._node_for(pid)
+        if node is not None:
+            from ..scheduling.requirements import Requirements
+
+            labels = ko.labels_of(node)
+            for req in Requirements.from_nodeclaim(nodeclaim):
+                if req.key in labels and not req.has(labels[req.key]):
+                    return DRIFT_REQUIREMENTS
         return ""
+
+    async def _node_for(self, provider_id: str):
+        for node in await self.instances.kube.list("v1", "Node"):
+            if ko.provider_id_of(node) == provider_id:
+                return node
+        return None
 
     def repair_policies(self) -> list:
         return self._repair_policies

@@ -272,3 +272,37 @@ def test_build_manager_registers_drift_controller():
         assert "DriftController" not in [type(c).__name__ for c in mgr2.controllers]
 
     run(main())
+
+
+def test_is_drifted_node_requirements_mutated():
+    """RequirementsDrift: a node label mutated out from under the claim's
+    requirements; keys absent from the node are NOT judged (platform may
+    simply not stamp them)."""
+    from gpu_provisioner_amd.cloudprovider.azure import DRIFT_REQUIREMENTS
+
+    async def main():
+        h = Harness().add_all_controllers(gc_interval=60.0, with_drift=False)
+        await h.start()
+        try:
+            nc = await provisioned(h, "reqd1")
+            node_name = nc["status"]["nodeName"]
+            node = await h.kube.get("v1", "Node", node_name)
+            # out-of-band: someone rewrote the instance-type label
+            await h.kube.patch(
+                "v1", "Node", node_name,
+                {"metadata": {"labels": {
+                    **node["metadata"]["labels"],
+                    karpv1.INSTANCE_TYPE_LABEL_KEY: "Standard_D4s_v5",
+                }}},
+            )
+            assert await h.cloud.is_drifted(nc) == DRIFT_REQUIREMENTS
+            # a requirement on a key the node does not carry is not drift
+            nc2 = await provisioned(h, "reqd2")
+            nc2["spec"]["requirements"].append(
+                {"key": "topology.kubernetes.io/zone", "operator": "In", "values": ["eastus2-1"]}
+            )
+            assert await h.cloud.is_drifted(nc2) == ""
+        finally:
+            await h.stop()
+
+    run(main())
