@@ -122,6 +122,10 @@ class AzureCloudProvider(CloudProvider):
         return ""
 
     async def _node_for(self, provider_id: str):
+        inf = getattr(self.instances, "nodes_informer", None)
+        if inf is not None and inf.has_synced and inf.has_index("providerID"):
+            nodes = inf.by_index("providerID", provider_id)
+            return nodes[0] if nodes else None
         for node in await self.instances.kube.list("v1", "Node"):
             if ko.provider_id_of(node) == provider_id:
                 return node
