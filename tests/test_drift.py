@@ -295,6 +295,17 @@ def test_is_drifted_node_requirements_mutated():
                     karpv1.INSTANCE_TYPE_LABEL_KEY: "Standard_D4s_v5",
                 }}},
             )
+            # drift reads through the informer cache: wait for the watch
+            # to deliver the mutation (production sweeps every 2 min)
+            async def informer_caught_up():
+                cached = h.nodes.get(node_name)
+                return (
+                    cached
+                    and ko.labels_of(cached).get(karpv1.INSTANCE_TYPE_LABEL_KEY)
+                    == "Standard_D4s_v5"
+                ) or None
+
+            await h.wait_for(informer_caught_up)
             assert await h.cloud.is_drifted(nc) == DRIFT_REQUIREMENTS
             # a requirement on a key the node does not carry is not drift
             nc2 = await provisioned(h, "reqd2")
