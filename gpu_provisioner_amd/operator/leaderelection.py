@@ -107,6 +107,24 @@ class LeaderElector:
                 return False
         return False
 
+    async def release(self) -> None:
+        """Graceful-shutdown lease hand-off: clear holderIdentity so a
+        follower acquires immediately instead of waiting out the lease
+        duration. Best-effort (the expiry path still covers crashes)."""
+        if not self.is_leader:
+            return
+        self.is_leader = False
+        try:
+            cur = await self.kube.get(
+                "coordination.k8s.io/v1", "Lease", self.name, self.namespace
+            )
+            if cur.get("spec", {}).get("holderIdentity") == self.identity:
+                cur["spec"]["holderIdentity"] = ""
+                await self.kube.update(cur)
+                log.info("leader election: released lease %s/%s", self.namespace, self.name)
+        except Exception as e:
+            log.warning("leader election: lease release failed (expiry will cover): %s", e)
+
     async def _renew(self) -> bool:
         try:
             cur = await self.kube.get("coordination.k8s.io/v1", "Lease", self.name, self.namespace)

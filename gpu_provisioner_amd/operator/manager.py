@@ -157,10 +157,27 @@ class Manager:
             await c.controller.stop()
 
     async def run_forever(self) -> None:
+        """Run until SIGTERM/SIGINT (what kubelet sends on pod stop), then
+        shut down gracefully: controllers stop, informers stop, and the
+        leader lease is RELEASED so a replica takes over immediately
+        instead of waiting out the lease duration."""
+        import signal
+
+        stop_event = asyncio.Event()
+        loop = asyncio.get_running_loop()
+        for sig in (signal.SIGTERM, signal.SIGINT):
+            try:
+                loop.add_signal_handler(sig, stop_event.set)
+            except (NotImplementedError, RuntimeError):
+                pass  # non-unix / nested-loop environments
         await self.start()
-        await asyncio.Event().wait()
+        await stop_event.wait()
+        log.info("shutdown signal received; stopping")
+        await self.stop()
 
     async def stop(self) -> None:
+        if self._elector is not None:
+            await self._elector.release()
         if self._elector_task is not None:
             self._elector_task.cancel()
             try:

@@ -51,7 +51,16 @@ def test_two_replicas_single_leader_and_failover():
             got = await h.wait_initialized("ha1", timeout=20)
             assert karpv1.is_initialized(got)
 
-            # leader dies abruptly (no graceful lease release)
+            # leader dies ABRUPTLY: kill the elector without releasing the
+            # lease (manager.stop() now releases gracefully — the crash
+            # path must still be covered by expiry takeover)
+            m1._elector.is_leader = False  # prevent release on stop
+            m1._elector_task.cancel()
+            try:
+                await m1._elector_task
+            except (asyncio.CancelledError, Exception):
+                pass
+            m1._elector_task = None
             await m1.stop()
             await wait_for(lambda: m2._elector.is_leader, timeout=20)
             assert all(c.controller._tasks for c in m2.controllers)
@@ -64,6 +73,44 @@ def test_two_replicas_single_leader_and_failover():
             await h.wait_gone(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "ha1", timeout=20)
             assert "ha1" not in h.agent_pools.pools
             assert "ha2" in h.agent_pools.pools
+        finally:
+            await m2.stop()
+
+    run(main(), timeout=120)
+
+
+def test_graceful_stop_releases_lease_for_fast_takeover():
+    """SIGTERM-path shutdown (manager.stop) must RELEASE the lease so the
+    follower acquires immediately — not after the lease duration."""
+    import time
+
+    async def main():
+        h = Harness(node_wait_interval=0.01)
+        m1 = build_manager(h.kube, leader_options(), h.cloud.inner, version="m1")
+        m2 = build_manager(h.kube, leader_options(), h.cloud.inner, version="m2")
+        for m in (m1, m2):
+            # LONG lease: only a released lease lets m2 take over quickly
+            m.lease_duration, m.renew_interval = 30.0, 0.3
+
+        async def wait_for(pred, timeout=15.0):
+            deadline = asyncio.get_event_loop().time() + timeout
+            while not pred():
+                if asyncio.get_event_loop().time() > deadline:
+                    raise TimeoutError
+                await asyncio.sleep(0.02)
+
+        await m1.start(serve_http=False)
+        await wait_for(lambda: m1._elector.is_leader)
+        await m2.start(serve_http=False)
+        await asyncio.sleep(0.5)
+        assert not m2._elector.is_leader
+        try:
+            t0 = time.monotonic()
+            await m1.stop()  # graceful: releases the lease
+            await wait_for(lambda: m2._elector.is_leader, timeout=10)
+            takeover = time.monotonic() - t0
+            # must be driven by the release, not the 30s lease expiry
+            assert takeover < 10, f"takeover took {takeover:.1f}s"
         finally:
             await m2.stop()
 
