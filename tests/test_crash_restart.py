@@ -170,3 +170,53 @@ def test_long_running_state_is_bounded_under_churn():
             await h.stop()
 
     run(main(), timeout=240)
+
+
+def test_restart_mid_chaos_converges():
+    """The hardest combination: controllers crash-restart WHILE ARM and
+    kube verbs are failing randomly and the fleet is mid-provision. The
+    fresh manager must still converge everything exactly-once."""
+    import random
+
+    from gpu_provisioner_amd.kube.client import ConflictError, TooManyRequestsError
+    from tests.test_chaos import ChaosError
+
+    async def main():
+        rng = random.Random(31337)
+        h = Harness(create_latency=0.2, node_wait_interval=0.01).add_all_controllers(**KW)
+        arm = [ChaosError(rng, p=0.08) for _ in range(4)]
+        (
+            h.agent_pools.create_error,
+            h.agent_pools.delete_error,
+            h.agent_pools.get_error,
+            h.agent_pools.list_error,
+        ) = arm
+
+        def kube_chaos(verb, gvk, payload):
+            if verb in ("update", "patch", "delete") and gvk[1] != "Pod":
+                if rng.random() < 0.06:
+                    return (
+                        ConflictError("chaos")
+                        if rng.random() < 0.5
+                        else TooManyRequestsError("chaos")
+                    )
+            return None
+
+        h.server.reactors.append(kube_chaos)
+        await h.start()
+        try:
+            names = [f"cr{i:02d}" for i in range(10)]
+            await asyncio.gather(*(h.kube.create(h.make_nodeclaim(n)) for n in names))
+            await asyncio.sleep(0.15)  # mid-LRO for most claims
+            await h.crash_restart_controllers(**KW)
+            await asyncio.sleep(0.2)
+            await h.crash_restart_controllers(**KW)  # twice, for spite
+            done = await asyncio.gather(
+                *(h.wait_initialized(n, timeout=240) for n in names)
+            )
+            assert all(karpv1.is_initialized(nc) for nc in done)
+            assert sorted(h.agent_pools.pools) == sorted(names)
+        finally:
+            await h.stop()
+
+    run(main(), timeout=400)
