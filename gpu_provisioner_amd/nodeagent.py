@@ -128,6 +128,12 @@ class NodeAgent:
         """v_mfma_f32_16x16x32_fp8_fp8 (E4M3) — the serving datapath."""
         return self.lib.na_mfma_fp8_selftest(dev) == 0
 
+    def mfma_bf16_tile_check(self, dev: int) -> bool:
+        """Layout-correct 16x16x32 bf16 GEMM tile with asymmetric data vs
+        an exact host reference — catches fragment-mapping errors the
+        uniform-operand self-tests cannot."""
+        return self.lib.na_mfma_bf16_tile_check(dev) == 0
+
     def lds_selftest(self, dev: int) -> tuple:
         """(ok, bytes_tested): whole-LDS pattern write/swizzled-read check."""
         tested = ctypes.c_longlong(0)
@@ -190,6 +196,8 @@ class NodeAgent:
                 g.mfma_fp8_ok = self.mfma_fp8_selftest(d)
                 if not g.mfma_fp8_ok:
                     g.problems.append(f"MFMA fp8 selftest failed: {self._err()}")
+                if not self.mfma_bf16_tile_check(d):
+                    g.problems.append(f"MFMA bf16 tile check failed: {self._err()}")
                 g.lds_ok, g.lds_bytes_tested = self.lds_selftest(d)
                 if not g.lds_ok:
                     g.problems.append(f"LDS selftest failed: {self._err()}")

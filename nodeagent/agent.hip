@@ -128,6 +128,58 @@ __global__ void mfma_fp8_selftest_kernel(float* __restrict__ out, long a_bits, l
 #endif
 }
 
+// Layout-correct bf16 GEMM tile: one v_mfma_f32_16x16x32_bf16 computing
+// D[16x16] = A[16x32]·B[32x16] with ASYMMETRIC integer-valued data, checked
+// against an exact host reference. The uniform-operand self-tests above are
+// layout-independent by construction; this one fails if the per-lane
+// fragment mapping (A: row=l&15, k=(l>>4)*8+i; B: col=l&15, same k;
+// D: col=l&15, row=(l>>4)*4+i) is wrong anywhere. Integer values keep both
+// sides exact regardless of summation order.
+__device__ __host__ inline float tile_a(int i, int k) { return (float)((i * 31 + k * 7) % 7 - 3); }
+__device__ __host__ inline float tile_b(int k, int j) { return (float)((k * 13 + j * 3) % 5 - 2); }
+
+__global__ void mfma_bf16_tile_kernel(float* __restrict__ out) {
+#if defined(__gfx950__)
+    int l = threadIdx.x;
+    bf16x8 a, b;
+    int arow = l & 15, kbase = (l >> 4) * 8;
+    for (int i = 0; i < 8; ++i) {
+        a[i] = (__bf16)tile_a(arow, kbase + i);
+        b[i] = (__bf16)tile_b(kbase + i, arow);  // B: col = l&15
+    }
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+    int dcol = l & 15, drow0 = (l >> 4) * 4;
+    for (int i = 0; i < 4; ++i) out[(drow0 + i) * 16 + dcol] = acc[i];
+#else
+    out[threadIdx.x] = -1.0f;
+#endif
+}
+
+extern "C" int na_mfma_bf16_tile_check(int dev) {
+    HIP_CHECK(hipSetDevice(dev));
+    float* out = nullptr;
+    HIP_CHECK(hipMalloc(&out, 256 * sizeof(float)));
+    mfma_bf16_tile_kernel<<<dim3(1), dim3(64)>>>(out);
+    HIP_CHECK(hipDeviceSynchronize());
+    float host[256];
+    HIP_CHECK(hipMemcpy(host, out, sizeof(host), hipMemcpyDeviceToHost));
+    (void)hipFree(out);
+    for (int i = 0; i < 16; ++i) {
+        for (int j = 0; j < 16; ++j) {
+            float ref = 0.f;
+            for (int k = 0; k < 32; ++k) ref += tile_a(i, k) * tile_b(k, j);
+            if (host[i * 16 + j] != ref) {
+                std::snprintf(na_last_error_buf, sizeof(na_last_error_buf),
+                              "mfma bf16 tile: D[%d][%d] got %g want %g", i, j,
+                              host[i * 16 + j], ref);
+                return NA_ERR_VERIFY;
+            }
+        }
+    }
+    return NA_OK;
+}
+
 // LDS self-test: fill the whole per-WG allocation with a position-dependent
 // pattern, barrier, read back through a bank-swizzled index. Exercises the
 // LDS array + crossbar across all CUs (one WG per CU's worth of a big grid).

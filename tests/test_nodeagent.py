@@ -36,6 +36,7 @@ def test_hip_library_builds_and_loads():
         "na_mfma_fp8_selftest",
         "na_lds_selftest",
         "na_sdma_bandwidth",
+        "na_mfma_bf16_tile_check",
         "na_p2p_matrix",
         "na_p2p_bandwidth",
         "na_last_error",
@@ -100,6 +101,9 @@ def test_gpu_mfma_datatype_paths():
     agent = NodeAgent()
     assert agent.mfma_bf16_selftest(0), "MFMA bf16 (v_mfma_f32_16x16x32_bf16) failed"
     assert agent.mfma_fp8_selftest(0), "MFMA fp8 (v_mfma_f32_16x16x32_fp8_fp8) failed"
+    # asymmetric-data GEMM tile vs exact host reference: catches fragment
+    # layout errors the uniform-operand checks cannot
+    assert agent.mfma_bf16_tile_check(0), "MFMA bf16 tile (layout) check failed"
 
 
 @pytest.mark.gpu
