@@ -134,6 +134,10 @@ class NodeAgent:
         uniform-operand self-tests cannot."""
         return self.lib.na_mfma_bf16_tile_check(dev) == 0
 
+    def mfma_fp8_tile_check(self, dev: int) -> bool:
+        """E4M3 variant of the layout-correct tile check."""
+        return self.lib.na_mfma_fp8_tile_check(dev) == 0
+
     def lds_selftest(self, dev: int) -> tuple:
         """(ok, bytes_tested): whole-LDS pattern write/swizzled-read check."""
         tested = ctypes.c_longlong(0)
@@ -198,6 +202,8 @@ class NodeAgent:
                     g.problems.append(f"MFMA fp8 selftest failed: {self._err()}")
                 if not self.mfma_bf16_tile_check(d):
                     g.problems.append(f"MFMA bf16 tile check failed: {self._err()}")
+                if not self.mfma_fp8_tile_check(d):
+                    g.problems.append(f"MFMA fp8 tile check failed: {self._err()}")
                 g.lds_ok, g.lds_bytes_tested = self.lds_selftest(d)
                 if not g.lds_ok:
                     g.problems.append(f"LDS selftest failed: {self._err()}")

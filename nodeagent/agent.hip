@@ -180,6 +180,67 @@ extern "C" int na_mfma_bf16_tile_check(int dev) {
     return NA_OK;
 }
 
+// fp8 (E4M3) variant of the layout-correct tile: same fragment maps, the
+// 8 per-lane elements packed one byte each into the i64 operand (element i
+// at byte i). Values restricted to small integers exact in E4M3.
+__device__ __host__ inline int tile8_a(int i, int k) { return (i * 31 + k * 7) % 7 - 3; }
+__device__ __host__ inline int tile8_b(int k, int j) { return (k * 13 + j * 3) % 5 - 2; }
+
+__device__ __host__ inline unsigned char e4m3(int v) {
+    // encodings for -3..3 (sign | exp(bias 7) | 3-bit mantissa)
+    switch (v) {
+        case 0: return 0x00;
+        case 1: return 0x38;
+        case 2: return 0x40;
+        case 3: return 0x44;
+        case -1: return 0xB8;
+        case -2: return 0xC0;
+        default: return 0xC4;  // -3
+    }
+}
+
+__global__ void mfma_fp8_tile_kernel(float* __restrict__ out) {
+#if defined(__gfx950__)
+    int l = threadIdx.x;
+    int arow = l & 15, kbase = (l >> 4) * 8;
+    long a_bits = 0, b_bits = 0;
+    for (int i = 0; i < 8; ++i) {
+        a_bits |= (long)e4m3(tile8_a(arow, kbase + i)) << (8 * i);
+        b_bits |= (long)e4m3(tile8_b(kbase + i, arow)) << (8 * i);
+    }
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a_bits, b_bits, acc, 0, 0, 0);
+    int dcol = l & 15, drow0 = (l >> 4) * 4;
+    for (int i = 0; i < 4; ++i) out[(drow0 + i) * 16 + dcol] = acc[i];
+#else
+    out[threadIdx.x] = -1.0f;
+#endif
+}
+
+extern "C" int na_mfma_fp8_tile_check(int dev) {
+    HIP_CHECK(hipSetDevice(dev));
+    float* out = nullptr;
+    HIP_CHECK(hipMalloc(&out, 256 * sizeof(float)));
+    mfma_fp8_tile_kernel<<<dim3(1), dim3(64)>>>(out);
+    HIP_CHECK(hipDeviceSynchronize());
+    float host[256];
+    HIP_CHECK(hipMemcpy(host, out, sizeof(host), hipMemcpyDeviceToHost));
+    (void)hipFree(out);
+    for (int i = 0; i < 16; ++i) {
+        for (int j = 0; j < 16; ++j) {
+            float ref = 0.f;
+            for (int k = 0; k < 32; ++k) ref += (float)(tile8_a(i, k) * tile8_b(k, j));
+            if (host[i * 16 + j] != ref) {
+                std::snprintf(na_last_error_buf, sizeof(na_last_error_buf),
+                              "mfma fp8 tile: D[%d][%d] got %g want %g", i, j,
+                              host[i * 16 + j], ref);
+                return NA_ERR_VERIFY;
+            }
+        }
+    }
+    return NA_OK;
+}
+
 // LDS self-test: fill the whole per-WG allocation with a position-dependent
 // pattern, barrier, read back through a bank-swizzled index. Exercises the
 // LDS array + crossbar across all CUs (one WG per CU's worth of a big grid).

@@ -37,6 +37,7 @@ def test_hip_library_builds_and_loads():
         "na_lds_selftest",
         "na_sdma_bandwidth",
         "na_mfma_bf16_tile_check",
+        "na_mfma_fp8_tile_check",
         "na_p2p_matrix",
         "na_p2p_bandwidth",
         "na_last_error",
@@ -104,6 +105,7 @@ def test_gpu_mfma_datatype_paths():
     # asymmetric-data GEMM tile vs exact host reference: catches fragment
     # layout errors the uniform-operand checks cannot
     assert agent.mfma_bf16_tile_check(0), "MFMA bf16 tile (layout) check failed"
+    assert agent.mfma_fp8_tile_check(0), "MFMA fp8 tile (layout) check failed"
 
 
 @pytest.mark.gpu
