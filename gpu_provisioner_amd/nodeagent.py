@@ -138,6 +138,10 @@ class NodeAgent:
         """E4M3 variant of the layout-correct tile check."""
         return self.lib.na_mfma_fp8_tile_check(dev) == 0
 
+    def mfma_i8_tile_check(self, dev: int) -> bool:
+        """int8 K=64 variant (v_mfma_i32_16x16x64_i8) of the tile check."""
+        return self.lib.na_mfma_i8_tile_check(dev) == 0
+
     def lds_selftest(self, dev: int) -> tuple:
         """(ok, bytes_tested): whole-LDS pattern write/swizzled-read check."""
         tested = ctypes.c_longlong(0)
@@ -204,6 +208,8 @@ class NodeAgent:
                     g.problems.append(f"MFMA bf16 tile check failed: {self._err()}")
                 if not self.mfma_fp8_tile_check(d):
                     g.problems.append(f"MFMA fp8 tile check failed: {self._err()}")
+                if not self.mfma_i8_tile_check(d):
+                    g.problems.append(f"MFMA i8 tile check failed: {self._err()}")
                 g.lds_ok, g.lds_bytes_tested = self.lds_selftest(d)
                 if not g.lds_ok:
                     g.problems.append(f"LDS selftest failed: {self._err()}")

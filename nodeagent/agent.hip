@@ -241,6 +241,54 @@ extern "C" int na_mfma_fp8_tile_check(int dev) {
     return NA_OK;
 }
 
+// i8 variant (v_mfma_i32_16x16x64_i8, K=64 — CDNA4's 2xK int8 path):
+// 16 per-lane elements packed 4-per-i32; k = (l>>4)*16 + i.
+typedef int i32x4 __attribute__((ext_vector_type(4)));
+
+__global__ void mfma_i8_tile_kernel(int* __restrict__ out) {
+#if defined(__gfx950__)
+    int l = threadIdx.x;
+    int arow = l & 15, kbase = (l >> 4) * 16;
+    i32x4 a = {0, 0, 0, 0}, b = {0, 0, 0, 0};
+    for (int i = 0; i < 16; ++i) {
+        unsigned char av = (unsigned char)(signed char)tile8_a(arow, kbase + i);
+        unsigned char bv = (unsigned char)(signed char)tile8_b(kbase + i, arow);
+        a[i / 4] |= (int)av << (8 * (i % 4));
+        b[i / 4] |= (int)bv << (8 * (i % 4));
+    }
+    i32x4 acc = {0, 0, 0, 0};
+    acc = __builtin_amdgcn_mfma_i32_16x16x64_i8(a, b, acc, 0, 0, 0);
+    int dcol = l & 15, drow0 = (l >> 4) * 4;
+    for (int i = 0; i < 4; ++i) out[(drow0 + i) * 16 + dcol] = acc[i];
+#else
+    out[threadIdx.x] = -1;
+#endif
+}
+
+extern "C" int na_mfma_i8_tile_check(int dev) {
+    HIP_CHECK(hipSetDevice(dev));
+    int* out = nullptr;
+    HIP_CHECK(hipMalloc(&out, 256 * sizeof(int)));
+    mfma_i8_tile_kernel<<<dim3(1), dim3(64)>>>(out);
+    HIP_CHECK(hipDeviceSynchronize());
+    int host[256];
+    HIP_CHECK(hipMemcpy(host, out, sizeof(host), hipMemcpyDeviceToHost));
+    (void)hipFree(out);
+    for (int i = 0; i < 16; ++i) {
+        for (int j = 0; j < 16; ++j) {
+            int ref = 0;
+            for (int k = 0; k < 64; ++k) ref += tile8_a(i, k) * tile8_b(k, j);
+            if (host[i * 16 + j] != ref) {
+                std::snprintf(na_last_error_buf, sizeof(na_last_error_buf),
+                              "mfma i8 tile: D[%d][%d] got %d want %d", i, j,
+                              host[i * 16 + j], ref);
+                return NA_ERR_VERIFY;
+            }
+        }
+    }
+    return NA_OK;
+}
+
 // LDS self-test: fill the whole per-WG allocation with a position-dependent
 // pattern, barrier, read back through a bank-swizzled index. Exercises the
 // LDS array + crossbar across all CUs (one WG per CU's worth of a big grid).
