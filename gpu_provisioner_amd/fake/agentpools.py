@@ -109,6 +109,9 @@ class FakeAgentPools(AgentPoolsAPI):
         self.delete_latency = delete_latency
         self.api_latency = api_latency
         self.create_error = ScriptedError()
+        # the LRO ends "Failed" AFTER accept — how real Azure surfaces most
+        # allocation failures (discovered during provisioning, not at PUT)
+        self.create_lro_error = ScriptedError()
         self.delete_error = ScriptedError()
         self.get_error = ScriptedError()
         self.list_error = ScriptedError()
@@ -140,6 +143,11 @@ class FakeAgentPools(AgentPoolsAPI):
             cur = self.pools.get(pool_name_)
             if cur is None:
                 raise ARMError(404, "NotFound", f"agent pool {pool_name_} was deleted mid-create")
+            try:
+                self.create_lro_error.check()
+            except Exception:
+                cur["properties"]["provisioningState"] = "Failed"
+                raise
             cur["properties"]["provisioningState"] = "Succeeded"
             if self.on_pool_ready is not None:
                 await self.on_pool_ready(cur)

@@ -333,3 +333,35 @@ def test_registration_liveness_gate_wiring():
     mgr2 = build_manager(h.kube, Options.from_env_and_args([], {}), h.cloud.inner)
     lc2 = [c for c in mgr2.controllers if isinstance(c, LifecycleController)][0]
     assert lc2.registration_ttl is None
+
+
+def test_lro_allocation_failure_deletes_claim_and_gc_cleans_failed_pool():
+    """Azure surfaces most allocation failures DURING the create LRO (the
+    PUT is accepted, the LRO ends Failed). Launch must map it to
+    InsufficientCapacity and delete the claim; the leaked Failed pool is
+    then instance-GC'd."""
+    from gpu_provisioner_amd.fake.harness import Harness
+    from gpu_provisioner_amd.providers.instance.armapi import ARMError
+
+    async def main():
+        h = Harness(node_wait_interval=0.01).add_all_controllers(
+            gc_interval=0.3, adoption_age=0.1
+        )
+        h.agent_pools.create_lro_error.set(
+            ARMError(200, "AllocationFailed", "zone exhausted mid-provision"),
+            max_calls=1,
+        )
+        await h.start()
+        try:
+            await h.kube.create(h.make_nodeclaim("lrofail1"))
+            # launch maps LRO failure → InsufficientCapacity → claim deleted
+            await h.wait_gone(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "lrofail1", timeout=30)
+
+            async def failed_pool_collected():
+                return "lrofail1" not in h.agent_pools.pools or None
+
+            await h.wait_for(failed_pool_collected, timeout=30)
+        finally:
+            await h.stop()
+
+    run(main())
