@@ -130,6 +130,17 @@ class Informer:
                 waiters.remove(entry)
             except ValueError:
                 pass
+            # drop empty registries: one leftover list per key/value ever
+            # waited on is a slow leak at churn rates
+            if not waiters:
+                if name:
+                    self._key_waiters.pop(key, None)
+                else:
+                    vmap = self._index_waiters.get(index)
+                    if vmap is not None:
+                        vmap.pop(value, None)
+                        if not vmap:
+                            self._index_waiters.pop(index, None)
 
     def _fire_waiters(self, waiters: Optional[list], event_type: str, obj: dict) -> None:
         if not waiters:

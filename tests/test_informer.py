@@ -202,3 +202,27 @@ def test_wait_until_by_name_and_index():
         await inf.stop()
 
     run(main())
+
+
+def test_wait_until_registries_do_not_leak():
+    from gpu_provisioner_amd.fake.apiserver import InMemoryAPIServer, InMemoryClient
+    from gpu_provisioner_amd.kube.informer import Informer
+
+    async def main():
+        kube = InMemoryClient(InMemoryAPIServer())
+        inf = Informer(kube, "v1", "Node")
+        inf.add_index("pool", lambda o: (o["metadata"].get("labels") or {}).get("agentpool"))
+        inf.start()
+        await inf.wait_for_sync()
+        for i in range(50):
+            with pytest.raises(asyncio.TimeoutError):
+                await inf.wait_until(lambda et, o: None, name=f"n{i}", timeout=0.001)
+            with pytest.raises(asyncio.TimeoutError):
+                await inf.wait_until(
+                    lambda et, o: None, index="pool", value=f"p{i}", timeout=0.001
+                )
+        assert not inf._key_waiters, inf._key_waiters
+        assert not inf._index_waiters, inf._index_waiters
+        await inf.stop()
+
+    run(main())
