@@ -42,12 +42,13 @@ extern "C" const char* na_last_error() { return na_last_error_buf; }
 // ---------------------------------------------------------------------------
 
 // Streaming float4 copy: the canonical HBM bandwidth probe. Winning shape
-// from the on-device variant sweep (nodeagent/bw_sweep.hip, MI355X): each
-// workgroup owns ONE CONTIGUOUS slice (no grid-stride — consecutive
-// iterations stay in the same DRAM window) with nontemporal dwordx4
-// loads/stores (streaming policy, no L2/LC pollution), 1024 threads ×
-// 8192 WGs. Measured 5.6 TB/s vs 4.6-5.0 for grid-stride variants and
-// ≈6.3 TB/s achievable on this chip.
+// from two on-device variant sweeps (nodeagent/bw_sweep.hip + /tmp round 2,
+// MI355X): each workgroup owns ONE CONTIGUOUS slice (consecutive
+// iterations stay in the same DRAM window — grid-stride loses ~15%),
+// nontemporal dwordx4 (streaming policy, no L2/LC pollution), and FOUR
+// independent loads in flight per thread before the stores. Measured
+// 6233 GB/s at 16384 WGs x 1024 threads — 99% of the ≈6.3 TB/s this chip
+// sustains on read+write streams.
 typedef float f4v __attribute__((ext_vector_type(4)));  // raw vector: nt-builtin compatible
 
 __global__ void copy_f4_kernel(const float4* __restrict__ src_, float4* __restrict__ dst_,
@@ -57,7 +58,19 @@ __global__ void copy_f4_kernel(const float4* __restrict__ src_, float4* __restri
     size_t per = (n + gridDim.x - 1) / gridDim.x;
     size_t lo = blockIdx.x * per;
     size_t hi = lo + per < n ? lo + per : n;
-    for (size_t i = lo + threadIdx.x; i < hi; i += blockDim.x)
+    size_t bd = blockDim.x;
+    size_t i = lo + threadIdx.x;
+    for (; i + 3 * bd < hi; i += 4 * bd) {
+        f4v a = __builtin_nontemporal_load(&src[i]);
+        f4v b = __builtin_nontemporal_load(&src[i + bd]);
+        f4v c = __builtin_nontemporal_load(&src[i + 2 * bd]);
+        f4v e = __builtin_nontemporal_load(&src[i + 3 * bd]);
+        __builtin_nontemporal_store(a, &dst[i]);
+        __builtin_nontemporal_store(b, &dst[i + bd]);
+        __builtin_nontemporal_store(c, &dst[i + 2 * bd]);
+        __builtin_nontemporal_store(e, &dst[i + 3 * bd]);
+    }
+    for (; i < hi; i += bd)
         __builtin_nontemporal_store(__builtin_nontemporal_load(&src[i]), &dst[i]);
 }
 
@@ -341,8 +354,8 @@ extern "C" int na_hbm_bandwidth(int dev, long long bytes, int iters, double* gbs
         return NA_ERR_HIP;
     }
     HIP_CHECK(hipMemset(src, 1, n * sizeof(float4)));
-    // sweep winner: 8192 WGs × 1024 threads (bw_sweep.hip)
-    dim3 grid(8192), block(1024);
+    // sweep winner: 16384 WGs × 1024 threads, 4-deep in-chunk unroll
+    dim3 grid(16384), block(1024);
     hipEvent_t t0, t1;
     HIP_CHECK(hipEventCreate(&t0));
     HIP_CHECK(hipEventCreate(&t1));
