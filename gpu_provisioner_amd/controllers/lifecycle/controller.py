@@ -120,10 +120,20 @@ class LifecycleController:
 
     async def reconcile(self, key: str) -> Optional[Result]:
         decorator.current_controller.set(self.NAME)
-        try:
-            nodeclaim = await self.kube.get(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, key)
-        except NotFoundError:
-            return None
+        # read from the informer cache (controller-runtime's cached-client
+        # default: no apiserver round-trip per reconcile); writes still carry
+        # resourceVersion preconditions, so a stale read just conflicts and
+        # requeues. Fall back to the apiserver before sync.
+        if self.nodeclaims.has_synced:
+            cached = self.nodeclaims.get(key)
+            if cached is None:
+                return None
+            nodeclaim = ko.deep_copy(cached)
+        else:
+            try:
+                nodeclaim = await self.kube.get(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, key)
+            except NotFoundError:
+                return None
         if not karpv1.is_managed(nodeclaim):
             return None
         if ko.is_deleting(nodeclaim):
@@ -184,6 +194,7 @@ class LifecycleController:
         if karpv1.is_launched(nodeclaim):
             return None
         uid = ko.uid_of(nodeclaim)
+        launched_now = False  # metrics/events only when WE created this pass
         cached = self._launch_cache.get(uid)
         if cached is not None and cached[0] > time.monotonic():
             created = cached[1]
@@ -212,6 +223,7 @@ class LifecycleController:
             # constant TTL ⇒ insertion order == expiry order: expire from
             # the front in O(1) (a full scan per insert cost 164µs at
             # steady-state churn — the cache holds rate×TTL entries)
+            launched_now = True
             self._launch_cache[uid] = (time.monotonic() + LAUNCH_CACHE_TTL, created)
             self._launch_cache.move_to_end(uid)
             nw = time.monotonic()
@@ -242,12 +254,16 @@ class LifecycleController:
                 status[f] = created_status[f]
         ko.set_condition(nodeclaim, karpv1.COND_LAUNCHED, ko.CONDITION_TRUE, "Launched")
         await self._patch_status(nodeclaim)
-        self._observe_since_creation(nodeclaim, LAUNCH_DURATION)
-        NODECLAIMS_CREATED.labels(
-            nodepool=ko.labels_of(nodeclaim).get(karpv1.NODEPOOL_LABEL_KEY, ""),
-            capacity_type=ko.labels_of(nodeclaim).get(karpv1.CAPACITY_TYPE_LABEL_KEY, ""),
-            instance_type=ko.labels_of(nodeclaim).get(karpv1.INSTANCE_TYPE_LABEL_KEY, ""),
-        ).inc()
+        if launched_now:
+            # a stale cached-client read can re-run this sub-reconciler after
+            # the claim is already Launched; the UID cache dedupes the cloud
+            # create, and this guard dedupes the bookkeeping
+            self._observe_since_creation(nodeclaim, LAUNCH_DURATION)
+            NODECLAIMS_CREATED.labels(
+                nodepool=ko.labels_of(nodeclaim).get(karpv1.NODEPOOL_LABEL_KEY, ""),
+                capacity_type=ko.labels_of(nodeclaim).get(karpv1.CAPACITY_TYPE_LABEL_KEY, ""),
+                instance_type=ko.labels_of(nodeclaim).get(karpv1.INSTANCE_TYPE_LABEL_KEY, ""),
+            ).inc()
         self.recorder.publish(nodeclaim, "Launched", f"instance {status['providerID']} launched")
         return None
 

@@ -123,10 +123,18 @@ class TerminationController:
 
     async def reconcile(self, key: str) -> Optional[Result]:
         decorator.current_controller.set(self.NAME)
-        try:
-            node = await self.kube.get("v1", "Node", key)
-        except NotFoundError:
-            return None
+        # cached-client read (see lifecycle.reconcile); stale reads conflict
+        # on write and requeue
+        if self.nodes.has_synced:
+            cached = self.nodes.get(key)
+            if cached is None:
+                return None
+            node = ko.deep_copy(cached)
+        else:
+            try:
+                node = await self.kube.get("v1", "Node", key)
+            except NotFoundError:
+                return None
         if not ko.is_deleting(node):
             return None
         if not ko.has_finalizer(node, karpv1.TERMINATION_FINALIZER):
