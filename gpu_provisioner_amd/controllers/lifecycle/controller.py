@@ -86,6 +86,11 @@ class LifecycleController:
         self.nodes = nodes
         # launch idempotency cache: uid -> (deadline, created nodeclaim-shaped dict)
         self._launch_cache: OrderedDict = OrderedDict()
+        # read-your-writes floor for cached-client reads: key -> highest rv
+        # THIS controller wrote. A cache older than our own last write must
+        # not be trusted — re-running the chain on pre-write state costs a
+        # stale pass per write (measured ~20% bench throughput).
+        self._written_rv: OrderedDict = OrderedDict()
         self.controller = Controller(
             self.NAME,
             self.reconcile,
@@ -102,7 +107,20 @@ class LifecycleController:
 
     # -- event mapping -------------------------------------------------------
 
+    def _record_write(self, name: str, obj: dict) -> None:
+        """Remember the rv of our own write (read-your-writes floor)."""
+        try:
+            rv = int(ko.meta(obj).get("resourceVersion") or 0)
+        except (TypeError, ValueError):
+            return
+        self._written_rv[name] = max(self._written_rv.get(name, 0), rv)
+        self._written_rv.move_to_end(name)
+        while len(self._written_rv) > 4096:
+            self._written_rv.popitem(last=False)
+
     def _on_nodeclaim_event(self, event_type: str, obj: dict) -> None:
+        if event_type == "DELETED":
+            self._written_rv.pop(ko.name_of(obj), None)
         if karpv1.is_managed(obj):
             self.controller.enqueue_nowait(ko.name_of(obj))
 
@@ -121,18 +139,23 @@ class LifecycleController:
     async def reconcile(self, key: str) -> Optional[Result]:
         decorator.current_controller.set(self.NAME)
         # read from the informer cache (controller-runtime's cached-client
-        # default: no apiserver round-trip per reconcile); writes still carry
-        # resourceVersion preconditions, so a stale read just conflicts and
-        # requeues. Fall back to the apiserver before sync.
+        # default: no apiserver round-trip per reconcile) — unless the cache
+        # is behind OUR OWN last write for this key (read-your-writes floor);
+        # writes still carry resourceVersion preconditions either way.
+        nodeclaim = None
         if self.nodeclaims.has_synced:
             cached = self.nodeclaims.get(key)
-            if cached is None:
-                return None
-            nodeclaim = ko.deep_copy(cached)
-        else:
+            if cached is not None:
+                rv = int(ko.meta(cached).get("resourceVersion") or 0)
+                if rv >= self._written_rv.get(key, 0):
+                    nodeclaim = ko.deep_copy(cached)
+            elif key not in self._written_rv:
+                return None  # cache authoritative: never seen / fully deleted
+        if nodeclaim is None:
             try:
                 nodeclaim = await self.kube.get(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, key)
             except NotFoundError:
+                self._written_rv.pop(key, None)
                 return None
         if not karpv1.is_managed(nodeclaim):
             return None
@@ -145,6 +168,7 @@ class LifecycleController:
             ko.add_finalizer(nodeclaim, karpv1.TERMINATION_FINALIZER)
             try:
                 nodeclaim = await self.kube.update(nodeclaim)
+                self._record_write(key, nodeclaim)
             except ConflictError:
                 return Result(requeue=True)
 
@@ -245,6 +269,7 @@ class LifecycleController:
             )
             ko.meta(nodeclaim)["labels"] = merged
             ko.meta(nodeclaim)["resourceVersion"] = updated["metadata"]["resourceVersion"]
+            self._record_write(ko.name_of(nodeclaim), updated)
         status = nodeclaim.setdefault("status", {})
         created_status = created.get("status", {})
         status["providerID"] = created_status.get("providerID", "")
@@ -450,7 +475,8 @@ class LifecycleController:
             pass  # instance gone — safe to release the NodeClaim
         ko.remove_finalizer(nodeclaim, karpv1.TERMINATION_FINALIZER)
         try:
-            await self.kube.update(nodeclaim)
+            updated = await self.kube.update(nodeclaim)
+            self._record_write(ko.name_of(nodeclaim), updated)
         except ConflictError:
             return Result(requeue=True)
         except NotFoundError:
@@ -539,6 +565,7 @@ class LifecycleController:
             # keep the in-hand object's resourceVersion fresh so a later
             # update (e.g. finalizer removal) doesn't conflict with our own write
             ko.meta(nodeclaim)["resourceVersion"] = updated["metadata"]["resourceVersion"]
+            self._record_write(ko.name_of(nodeclaim), updated)
             return
         log.warning(
             "NodeClaim %s: status patch abandoned after repeated conflicts",
