@@ -82,6 +82,10 @@ class TerminationController:
             # reference scales 100-5000 with CPU (termination/controller.go:58-61)
             workers=workers if workers is not None else linear_scale_reconciles(32, 512),
         )
+        from collections import OrderedDict
+        # read-your-writes floor (see lifecycle): node-cache reads are
+        # trusted only at-or-above our own last write's rv
+        self._written_rv: OrderedDict = OrderedDict()
         nodeclaims.add_index(
             "providerID", lambda o: o.get("status", {}).get("providerID") or None
         )
@@ -100,7 +104,19 @@ class TerminationController:
         if volumeattachments is not None:
             volumeattachments.add_handler(self._on_va_event)
 
+    def _record_write(self, name: str, obj: dict) -> None:
+        try:
+            rv = int(obj.get("metadata", {}).get("resourceVersion") or 0)
+        except (TypeError, ValueError):
+            return
+        self._written_rv[name] = max(self._written_rv.get(name, 0), rv)
+        self._written_rv.move_to_end(name)
+        while len(self._written_rv) > 4096:
+            self._written_rv.popitem(last=False)
+
     def _on_node_event(self, event_type: str, obj: dict) -> None:
+        if event_type == "DELETED":
+            self._written_rv.pop(ko.name_of(obj), None)
         if ko.is_deleting(obj) and ko.has_finalizer(obj, karpv1.TERMINATION_FINALIZER):
             self.controller.enqueue_nowait(ko.name_of(obj))
 
@@ -123,17 +139,21 @@ class TerminationController:
 
     async def reconcile(self, key: str) -> Optional[Result]:
         decorator.current_controller.set(self.NAME)
-        # cached-client read (see lifecycle.reconcile); stale reads conflict
-        # on write and requeue
+        # cached-client read with a read-your-writes floor (see lifecycle)
+        node = None
         if self.nodes.has_synced:
             cached = self.nodes.get(key)
-            if cached is None:
+            if cached is not None:
+                rv = int(cached.get("metadata", {}).get("resourceVersion") or 0)
+                if rv >= self._written_rv.get(key, 0):
+                    node = ko.deep_copy(cached)
+            elif key not in self._written_rv:
                 return None
-            node = ko.deep_copy(cached)
-        else:
+        if node is None:
             try:
                 node = await self.kube.get("v1", "Node", key)
             except NotFoundError:
+                self._written_rv.pop(key, None)
                 return None
         if not ko.is_deleting(node):
             return None
@@ -256,7 +276,8 @@ class TerminationController:
         if not ko.remove_finalizer(node, karpv1.TERMINATION_FINALIZER):
             return None
         try:
-            await self.kube.update(node)
+            updated = await self.kube.update(node)
+            self._record_write(ko.name_of(node), updated)
         except NotFoundError:
             return None
         except ConflictError:
@@ -298,7 +319,12 @@ class TerminationController:
                 **labels,
                 karpv1.EXCLUDE_FROM_LB_LABEL_KEY: "karpenter",
             }
-        await self.kube.patch("v1", "Node", ko.name_of(node), patch)
+        updated = await self.kube.patch("v1", "Node", ko.name_of(node), patch)
+        self._record_write(ko.name_of(node), updated)
+        # sync the in-hand node so later writes (finalizer removal) carry the
+        # post-taint rv instead of conflicting against our own patch
+        node.clear()
+        node.update(updated)
 
     async def _nodeclaim_for_node(self, node: dict) -> Optional[dict]:
         pid = ko.provider_id_of(node)
