@@ -199,3 +199,36 @@ def test_manager_probes_and_metrics_endpoints():
         await informers.stop_all()
 
     run(main())
+
+
+def test_profiling_debug_endpoints():
+    """ENABLE_PROFILING exposes pprof-equivalent debug routes on the
+    metrics app (reference operator.go:181-197)."""
+    from gpu_provisioner_amd.fake.apiserver import InMemoryAPIServer, InMemoryClient
+    from gpu_provisioner_amd.operator.manager import Manager
+    from gpu_provisioner_amd.operator.options import Options
+
+    async def main():
+        kube = InMemoryClient(InMemoryAPIServer())
+        opts = Options.from_env_and_args(["--enable-profiling"], {})
+        mgr = Manager(kube, opts)
+        app = httpx.AsyncClient(
+            transport=httpx.ASGITransport(app=mgr._metrics_app()), base_url="http://t"
+        )
+        resp = await app.get("/metrics")
+        assert resp.status_code == 200 and b"karpenter" in resp.content
+        resp = await app.get("/debug/pprof/goroutine")
+        assert resp.status_code == 200 and resp.json()  # per-thread stacks
+        resp = await app.get("/debug/tasks")
+        assert resp.status_code == 200
+        # without the flag the debug routes are absent
+        mgr2 = Manager(kube, Options.from_env_and_args([], {}))
+        app2 = httpx.AsyncClient(
+            transport=httpx.ASGITransport(app=mgr2._metrics_app()), base_url="http://t"
+        )
+        resp = await app2.get("/debug/tasks")
+        assert resp.status_code == 404
+        await app.aclose()
+        await app2.aclose()
+
+    run(main())
