@@ -157,6 +157,11 @@ class InMemoryAPIServer:
             m["generation"] = 1
             self._bump(stored)
             self._store[gvk][key] = stored
+            # Events are capped like a real cluster's event TTL would bound
+            # them — long soaks otherwise grow the store without limit
+            if kind == "Event" and len(self._store[gvk]) > 20000:
+                drop = next(iter(self._store[gvk]))
+                del self._store[gvk][drop]
             self._broadcast(gvk, ADDED, stored)
             return ko.deep_copy(stored)
 
