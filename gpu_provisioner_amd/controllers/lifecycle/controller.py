@@ -10,9 +10,13 @@ error taxonomy handling :79-124, status population :126-140), registration.go
 gating on amd.com/gpu, the AMD device plugin's resource).
 
 Design differences from the reference (deliberate):
-  * reads go through the client (read-your-writes on the in-memory path and
-    resourceVersion-fresh on HTTP), so the reference's 1 s post-patch sleep
-    (controller.go:160-173) is unnecessary — provision p50 improves by ~1 s;
+  * reconcile reads come from the informer cache (controller-runtime's
+    cached-client semantics) guarded by a read-your-writes floor — the
+    cache is only trusted at-or-above the rv of this controller's own
+    last write, else one fresh apiserver GET. That gives cache-speed
+    foreign-event reconciles without stale re-runs after our own writes,
+    and makes the reference's 1 s post-patch sleep (controller.go:160-173)
+    unnecessary — provision p50 improves by ~1 s;
   * the liveness sub-reconciler (delete NodeClaims that never Register
     within a timeout) is IMPLEMENTED but off by default: the reference
     disabled it in-tree (controller.go:154) because agent-pool creates can
