@@ -142,6 +142,10 @@ class NodeAgent:
         """int8 K=64 variant (v_mfma_i32_16x16x64_i8) of the tile check."""
         return self.lib.na_mfma_i8_tile_check(dev) == 0
 
+    def mfma_f16_tile_check(self, dev: int) -> bool:
+        """f16 variant (v_mfma_f32_16x16x32_f16) of the tile check."""
+        return self.lib.na_mfma_f16_tile_check(dev) == 0
+
     def lds_selftest(self, dev: int) -> tuple:
         """(ok, bytes_tested): whole-LDS pattern write/swizzled-read check."""
         tested = ctypes.c_longlong(0)
@@ -210,6 +214,8 @@ class NodeAgent:
                     g.problems.append(f"MFMA fp8 tile check failed: {self._err()}")
                 if not self.mfma_i8_tile_check(d):
                     g.problems.append(f"MFMA i8 tile check failed: {self._err()}")
+                if not self.mfma_f16_tile_check(d):
+                    g.problems.append(f"MFMA f16 tile check failed: {self._err()}")
                 g.lds_ok, g.lds_bytes_tested = self.lds_selftest(d)
                 if not g.lds_ok:
                     g.problems.append(f"LDS selftest failed: {self._err()}")

@@ -193,6 +193,51 @@ extern "C" int na_mfma_bf16_tile_check(int dev) {
     return NA_OK;
 }
 
+// f16 variant of the layout-correct tile (same fragment maps as bf16).
+typedef _Float16 f16x8 __attribute__((ext_vector_type(8)));
+
+__global__ void mfma_f16_tile_kernel(float* __restrict__ out) {
+#if defined(__gfx950__)
+    int l = threadIdx.x;
+    int arow = l & 15, kbase = (l >> 4) * 8;
+    f16x8 a, b;
+    for (int i = 0; i < 8; ++i) {
+        a[i] = (_Float16)tile_a(arow, kbase + i);
+        b[i] = (_Float16)tile_b(kbase + i, arow);
+    }
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_f16(a, b, acc, 0, 0, 0);
+    int dcol = l & 15, drow0 = (l >> 4) * 4;
+    for (int i = 0; i < 4; ++i) out[(drow0 + i) * 16 + dcol] = acc[i];
+#else
+    out[threadIdx.x] = -1.0f;
+#endif
+}
+
+extern "C" int na_mfma_f16_tile_check(int dev) {
+    HIP_CHECK(hipSetDevice(dev));
+    float* out = nullptr;
+    HIP_CHECK(hipMalloc(&out, 256 * sizeof(float)));
+    mfma_f16_tile_kernel<<<dim3(1), dim3(64)>>>(out);
+    HIP_CHECK(hipDeviceSynchronize());
+    float host[256];
+    HIP_CHECK(hipMemcpy(host, out, sizeof(host), hipMemcpyDeviceToHost));
+    (void)hipFree(out);
+    for (int i = 0; i < 16; ++i) {
+        for (int j = 0; j < 16; ++j) {
+            float ref = 0.f;
+            for (int k = 0; k < 32; ++k) ref += tile_a(i, k) * tile_b(k, j);
+            if (host[i * 16 + j] != ref) {
+                std::snprintf(na_last_error_buf, sizeof(na_last_error_buf),
+                              "mfma f16 tile: D[%d][%d] got %g want %g", i, j,
+                              host[i * 16 + j], ref);
+                return NA_ERR_VERIFY;
+            }
+        }
+    }
+    return NA_OK;
+}
+
 // fp8 (E4M3) variant of the layout-correct tile: same fragment maps, the
 // 8 per-lane elements packed one byte each into the i64 operand (element i
 // at byte i). Values restricted to small integers exact in E4M3.
