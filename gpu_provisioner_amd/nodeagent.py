@@ -146,6 +146,12 @@ class NodeAgent:
         """f16 variant (v_mfma_f32_16x16x32_f16) of the tile check."""
         return self.lib.na_mfma_f16_tile_check(dev) == 0
 
+    def mfma_mx_tile_check(self, dev: int) -> bool:
+        """Block-scaled MX path (v_mfma_scale_f32_16x16x128_f8f6f4, the
+        fp4/fp6/MX serving pipe): layout-correct fp8 data + E8M0 scale
+        semantics vs exact host references."""
+        return self.lib.na_mfma_mx_tile_check(dev) == 0
+
     def lds_selftest(self, dev: int) -> tuple:
         """(ok, bytes_tested): whole-LDS pattern write/swizzled-read check."""
         tested = ctypes.c_longlong(0)
@@ -216,6 +222,8 @@ class NodeAgent:
                     g.problems.append(f"MFMA i8 tile check failed: {self._err()}")
                 if not self.mfma_f16_tile_check(d):
                     g.problems.append(f"MFMA f16 tile check failed: {self._err()}")
+                if not self.mfma_mx_tile_check(d):
+                    g.problems.append(f"MFMA MX-scaled tile check failed: {self._err()}")
                 g.lds_ok, g.lds_bytes_tested = self.lds_selftest(d)
                 if not g.lds_ok:
                     g.problems.append(f"LDS selftest failed: {self._err()}")

@@ -347,6 +347,67 @@ extern "C" int na_mfma_i8_tile_check(int dev) {
     return NA_OK;
 }
 
+// Block-scaled MX path (gfx950-only): v_mfma_scale_f32_16x16x128_f8f6f4 —
+// the ONLY large-K low-precision MFMA and the fp4/fp6/MX serving pipe.
+// Two checks in one: (1) layout-correct asymmetric fp8-e4m3 data at
+// scale 1.0 against an exact host reference (per-lane A: row=l&15,
+// k=(l>>4)*32+i, byte i of <8 x i32>; B: col=l&15, same k; C/D as the
+// 16x16 family); (2) scale semantics — E8M0 exponent 128 on A doubles
+// the result.
+typedef int i32x8 __attribute__((ext_vector_type(8)));
+
+__global__ void mfma_mx_tile_kernel(float* __restrict__ out, int scale_a) {
+#if defined(__gfx950__)
+    int l = threadIdx.x;
+    int arow = l & 15, kbase = (l >> 4) * 32;
+    i32x8 a = {0, 0, 0, 0, 0, 0, 0, 0}, b = {0, 0, 0, 0, 0, 0, 0, 0};
+    for (int i = 0; i < 32; ++i) {
+        a[i / 4] |= (int)e4m3(tile8_a(arow, (kbase + i) % 64)) << (8 * (i % 4));
+        b[i / 4] |= (int)e4m3(tile8_b((kbase + i) % 64, arow)) << (8 * (i % 4));
+    }
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    int sb = 0x7F7F7F7F;  // E8M0 127 = 1.0 per block
+    int sa = scale_a * 0x01010101;
+    acc = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+        a, b, acc, /*cbsz=*/0, /*blgp=*/0, /*opselA=*/0, sa, /*opselB=*/0, sb);
+    int dcol = l & 15, drow0 = (l >> 4) * 4;
+    for (int i = 0; i < 4; ++i) out[(drow0 + i) * 16 + dcol] = acc[i];
+#else
+    out[threadIdx.x] = -1.0f;
+#endif
+}
+
+extern "C" int na_mfma_mx_tile_check(int dev) {
+    HIP_CHECK(hipSetDevice(dev));
+    float* out = nullptr;
+    HIP_CHECK(hipMalloc(&out, 256 * sizeof(float)));
+    for (int pass = 0; pass < 2; ++pass) {
+        int scale_a = pass == 0 ? 0x7F : 0x80;  // 1.0, then 2.0
+        float mult = pass == 0 ? 1.0f : 2.0f;
+        mfma_mx_tile_kernel<<<dim3(1), dim3(64)>>>(out, scale_a);
+        HIP_CHECK(hipDeviceSynchronize());
+        float host[256];
+        HIP_CHECK(hipMemcpy(host, out, sizeof(host), hipMemcpyDeviceToHost));
+        for (int i = 0; i < 16; ++i) {
+            for (int j = 0; j < 16; ++j) {
+                float ref = 0.f;
+                for (int k = 0; k < 128; ++k)
+                    ref += (float)(tile8_a(i, k % 64) * tile8_b(k % 64, j));
+                ref *= mult;
+                if (host[i * 16 + j] != ref) {
+                    std::snprintf(na_last_error_buf, sizeof(na_last_error_buf),
+                                  "mfma mx tile (scale %.0fx): D[%d][%d] got %g want %g",
+                                  mult, i, j, host[i * 16 + j], ref);
+                    (void)hipFree(out);
+                    return NA_ERR_VERIFY;
+                }
+            }
+        }
+    }
+    (void)hipFree(out);
+    return NA_OK;
+}
+
 // LDS self-test: fill the whole per-WG allocation with a position-dependent
 // pattern, barrier, read back through a bank-swizzled index. Exercises the
 // LDS array + crossbar across all CUs (one WG per CU's worth of a big grid).

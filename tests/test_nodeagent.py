@@ -40,6 +40,7 @@ def test_hip_library_builds_and_loads():
         "na_mfma_fp8_tile_check",
         "na_mfma_i8_tile_check",
         "na_mfma_f16_tile_check",
+        "na_mfma_mx_tile_check",
         "na_p2p_matrix",
         "na_p2p_bandwidth",
         "na_last_error",
@@ -110,6 +111,7 @@ def test_gpu_mfma_datatype_paths():
     assert agent.mfma_fp8_tile_check(0), "MFMA fp8 tile (layout) check failed"
     assert agent.mfma_i8_tile_check(0), "MFMA i8 tile (layout) check failed"
     assert agent.mfma_f16_tile_check(0), "MFMA f16 tile (layout) check failed"
+    assert agent.mfma_mx_tile_check(0), "MFMA MX-scaled tile (layout+scale) check failed"
 
 
 @pytest.mark.gpu
