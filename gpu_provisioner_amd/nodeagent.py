@@ -146,6 +146,11 @@ class NodeAgent:
         """f16 variant (v_mfma_f32_16x16x32_f16) of the tile check."""
         return self.lib.na_mfma_f16_tile_check(dev) == 0
 
+    def mfma_bf16_tile32_check(self, dev: int) -> bool:
+        """32x32x16 bf16 tile — the other CDNA4 tile geometry, with its
+        distinct C/D fragment map."""
+        return self.lib.na_mfma_bf16_tile32_check(dev) == 0
+
     def mfma_mx_tile_check(self, dev: int) -> bool:
         """Block-scaled MX path (v_mfma_scale_f32_16x16x128_f8f6f4, the
         fp4/fp6/MX serving pipe): layout-correct fp8 data + E8M0 scale
@@ -224,6 +229,8 @@ class NodeAgent:
                     g.problems.append(f"MFMA f16 tile check failed: {self._err()}")
                 if not self.mfma_mx_tile_check(d):
                     g.problems.append(f"MFMA MX-scaled tile check failed: {self._err()}")
+                if not self.mfma_bf16_tile32_check(d):
+                    g.problems.append(f"MFMA 32x32 tile check failed: {self._err()}")
                 g.lds_ok, g.lds_bytes_tested = self.lds_selftest(d)
                 if not g.lds_ok:
                     g.problems.append(f"LDS selftest failed: {self._err()}")

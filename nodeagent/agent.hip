@@ -347,6 +347,58 @@ extern "C" int na_mfma_i8_tile_check(int dev) {
     return NA_OK;
 }
 
+// 32x32 geometry: v_mfma_f32_32x32x16_bf16 — same pipes, the OTHER tile
+// shape with its own fragment maps (A: row=l&31, k=(l>>5)*8+i; B: col
+// likewise; C/D per the ISA: col=lane&31, row=(reg&3)+8*(reg>>2)+4*(lane>>5),
+// 16 accumulator elements per lane).
+typedef float f32x16 __attribute__((ext_vector_type(16)));
+
+__global__ void mfma_bf16_tile32_kernel(float* __restrict__ out) {
+#if defined(__gfx950__)
+    int l = threadIdx.x;
+    int arow = l & 31, kbase = (l >> 5) * 8;
+    bf16x8 a, b;
+    for (int i = 0; i < 8; ++i) {
+        a[i] = (__bf16)tile_a(arow, kbase + i);
+        b[i] = (__bf16)tile_b(kbase + i, arow);
+    }
+    f32x16 acc;
+    for (int i = 0; i < 16; ++i) acc[i] = 0.f;
+    acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc, 0, 0, 0);
+    int dcol = l & 31;
+    for (int r = 0; r < 16; ++r) {
+        int drow = (r & 3) + 8 * (r >> 2) + 4 * (l >> 5);
+        out[drow * 32 + dcol] = acc[r];
+    }
+#else
+    out[threadIdx.x] = -1.0f;
+#endif
+}
+
+extern "C" int na_mfma_bf16_tile32_check(int dev) {
+    HIP_CHECK(hipSetDevice(dev));
+    float* out = nullptr;
+    HIP_CHECK(hipMalloc(&out, 1024 * sizeof(float)));
+    mfma_bf16_tile32_kernel<<<dim3(1), dim3(64)>>>(out);
+    HIP_CHECK(hipDeviceSynchronize());
+    float host[1024];
+    HIP_CHECK(hipMemcpy(host, out, sizeof(host), hipMemcpyDeviceToHost));
+    (void)hipFree(out);
+    for (int i = 0; i < 32; ++i) {
+        for (int j = 0; j < 32; ++j) {
+            float ref = 0.f;
+            for (int k = 0; k < 16; ++k) ref += tile_a(i, k) * tile_b(k, j);
+            if (host[i * 32 + j] != ref) {
+                std::snprintf(na_last_error_buf, sizeof(na_last_error_buf),
+                              "mfma bf16 32x32 tile: D[%d][%d] got %g want %g", i, j,
+                              host[i * 32 + j], ref);
+                return NA_ERR_VERIFY;
+            }
+        }
+    }
+    return NA_OK;
+}
+
 // Block-scaled MX path (gfx950-only): v_mfma_scale_f32_16x16x128_f8f6f4 —
 // the ONLY large-K low-precision MFMA and the fp4/fp6/MX serving pipe.
 // Two checks in one: (1) layout-correct asymmetric fp8-e4m3 data at

@@ -41,6 +41,7 @@ def test_hip_library_builds_and_loads():
         "na_mfma_i8_tile_check",
         "na_mfma_f16_tile_check",
         "na_mfma_mx_tile_check",
+        "na_mfma_bf16_tile32_check",
         "na_p2p_matrix",
         "na_p2p_bandwidth",
         "na_last_error",
@@ -112,6 +113,7 @@ def test_gpu_mfma_datatype_paths():
     assert agent.mfma_i8_tile_check(0), "MFMA i8 tile (layout) check failed"
     assert agent.mfma_f16_tile_check(0), "MFMA f16 tile (layout) check failed"
     assert agent.mfma_mx_tile_check(0), "MFMA MX-scaled tile (layout+scale) check failed"
+    assert agent.mfma_bf16_tile32_check(0), "MFMA 32x32 tile (layout) check failed"
 
 
 @pytest.mark.gpu
