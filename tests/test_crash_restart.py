@@ -220,3 +220,32 @@ def test_restart_mid_chaos_converges():
             await h.stop()
 
     run(main(), timeout=400)
+
+
+def test_idle_controller_does_not_busy_poll():
+    """A resting controller (one Initialized claim, production cadences)
+    must consume ~zero CPU — guards against requeue-storm regressions of
+    the busy-poll class (measured 0.01% on the fixed code; livelocked
+    versions burned a full core)."""
+    import resource
+    import time
+
+    async def main():
+        h = Harness().add_all_controllers(gc_interval=120.0, drift_interval=120.0)
+        await h.start()
+        try:
+            await h.kube.create(h.make_nodeclaim("idle1"))
+            await h.wait_initialized("idle1")
+            await asyncio.sleep(0.5)  # settle
+            c0 = resource.getrusage(resource.RUSAGE_SELF)
+            t0 = time.monotonic()
+            await asyncio.sleep(3.0)
+            c1 = resource.getrusage(resource.RUSAGE_SELF)
+            wall = time.monotonic() - t0
+            cpu = (c1.ru_utime + c1.ru_stime) - (c0.ru_utime + c0.ru_stime)
+            # generous bound for noisy CI machines; busy-polling is >50%
+            assert cpu / wall < 0.05, f"idle CPU {cpu/wall*100:.1f}%"
+        finally:
+            await h.stop()
+
+    run(main())
