@@ -324,3 +324,64 @@ def test_full_topology_chaos_with_repair_and_drift():
             await h.stop()
 
     run(main(), timeout=400)
+
+
+def test_delete_recreate_same_name_while_watches_broken():
+    """A claim is deleted AND recreated (same name, new uid) while every
+    watch stream is down: the relist coalesces the transition into a
+    MODIFIED for the same key. The recreated claim must converge — the
+    old pool (same agent-pool name) is either adopted or GC'd and
+    recreated, and the final state belongs to the NEW uid."""
+
+    async def main():
+        h = Harness(node_wait_interval=0.005).add_all_controllers(
+            gc_interval=0.3, adoption_age=0.2, termination_requeue=0.01,
+            drain_requeue=0.01, instance_poll=0.01,
+        )
+        await h.start()
+        try:
+            await h.kube.create(h.make_nodeclaim("phoenix"))
+            first = await h.wait_initialized("phoenix")
+            old_uid = first["metadata"]["uid"]
+
+            h.server.break_watches()
+            # while informers relist: tear down and recreate under the
+            # same name (finalizer flow still runs server-side)
+            await h.kube.delete(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "phoenix")
+
+            async def old_gone():
+                try:
+                    nc = await h.kube.get(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "phoenix")
+                    return None if nc["metadata"]["uid"] == old_uid else True
+                except Exception:
+                    return True
+
+            await h.wait_for(old_gone, timeout=30)
+            h.server.break_watches()
+            await h.kube.create(h.make_nodeclaim("phoenix"))
+            h.server.break_watches()
+
+            async def new_ready():
+                try:
+                    nc = await h.kube.get(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "phoenix")
+                except Exception:
+                    return None
+                if nc["metadata"]["uid"] == old_uid:
+                    return None
+                return nc if karpv1.is_initialized(nc) else None
+
+            fresh = await h.wait_for(new_ready, timeout=60, interval=0.05)
+            assert fresh["metadata"]["uid"] != old_uid
+            assert "phoenix" in h.agent_pools.pools
+            # exactly one claim and its pool; no zombie node from the old life
+            claims = await h.kube.list(karpv1.API_VERSION, karpv1.KIND_NODECLAIM)
+            assert len(claims) == 1
+            nodes = [
+                n for n in await h.kube.list("v1", "Node")
+                if ko.labels_of(n).get("agentpool") == "phoenix"
+            ]
+            assert len(nodes) == 1, [ko.name_of(n) for n in nodes]
+        finally:
+            await h.stop()
+
+    run(main(), timeout=240)
