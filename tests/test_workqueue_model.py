@@ -70,7 +70,7 @@ def test_no_lost_work_randomized():
         await asyncio.gather(*workers)
 
     async def main():
-        for seed in range(25):
+        for seed in range(40):
             await one_seed(seed)
 
     run(main(), timeout=300)
