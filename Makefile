@@ -37,6 +37,13 @@ gpu-test: nodeagent
 e2etests:
 	$(PYTHON) -m pytest tests/test_e2e_suite.py -q
 
+# reliability tiers: chaos (CHAOS_SEED=N sweeps interleavings), crash-restart,
+# HA failover, workqueue/informer model checks
+chaos-test:
+	$(PYTHON) -m pytest tests/test_chaos.py tests/test_crash_restart.py \
+	  tests/test_ha_failover.py tests/test_workqueue_model.py \
+	  tests/test_informer_consistency.py -q
+
 bench:
 	$(PYTHON) bench.py --steps 10 --warmup 3
 
