@@ -72,6 +72,9 @@ class Options:
             return env.get(key, default)
 
         p = argparse.ArgumentParser("gpu-provisioner-amd")
+        from .. import __version__
+
+        p.add_argument("--version", action="version", version=f"gpu-provisioner-amd {__version__}")
         p.add_argument("--karpenter-service", default=envv("KARPENTER_SERVICE", ""))
         p.add_argument("--metrics-port", type=int, default=int(envv("METRICS_PORT", "8080")))
         p.add_argument(
