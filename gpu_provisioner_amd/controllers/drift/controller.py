@@ -150,5 +150,8 @@ class DriftController:
                 name,
                 uid_precondition=ko.uid_of(nodeclaim),
             )
-        except NotFoundError:
+        except (NotFoundError, ConflictError):
+            # NotFound: already gone. Conflict: uid changed under the same
+            # name — the claim was already replaced; either way nothing to
+            # do, and the exception must not abort the sweep's gather.
             pass

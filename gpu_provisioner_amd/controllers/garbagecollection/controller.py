@@ -48,11 +48,36 @@ class InstanceGCController:
         recorder: EventRecorder,
         interval: float = GC_INTERVAL,
         adoption_age: float = ADOPTION_AGE_SECONDS,
+        nodes=None,  # Optional[Informer]: indexed leaked-Node lookups
     ):
         self.kube = kube
         self.cloud = cloud
         self.recorder = recorder
         self.adoption_age = adoption_age
+        self.nodes = nodes
+        if nodes is not None:
+            nodes.add_index(
+                "providerID", lambda o: o.get("spec", {}).get("providerID") or None
+            )
+            # same fn shape as InstanceProvider.set_nodes_informer — add_index
+            # is first-registration-wins by name, so both must index BOTH
+            # agentpool label keys
+            nodes.add_index(
+                "agentpool",
+                lambda o: [
+                    v
+                    for v in (
+                        (o.get("metadata", {}).get("labels") or {}).get(
+                            karpv1.AGENTPOOL_LABEL_KEY
+                        ),
+                        (o.get("metadata", {}).get("labels") or {}).get(
+                            karpv1.AZURE_AGENTPOOL_LABEL_KEY
+                        ),
+                    )
+                    if v
+                ]
+                or None,
+            )
         self.controller = SingletonController(self.NAME, self.reconcile, interval)
 
     async def reconcile(self, key: str) -> Optional[Result]:
@@ -106,18 +131,31 @@ class InstanceGCController:
     async def _delete_leaked_nodes(self, inst: dict) -> None:
         pid = karpv1.provider_id_of(inst)
         name = ko.name_of(inst)
-        for node in await self.kube.list("v1", "Node"):
-            if (
-                ko.provider_id_of(node) == pid and pid
-            ) or ko.labels_of(node).get(karpv1.AGENTPOOL_LABEL_KEY) == name:
-                try:
-                    # strip our finalizer first: the instance is already gone,
-                    # there is nothing left to drain
-                    if ko.remove_finalizer(node, karpv1.TERMINATION_FINALIZER):
-                        await self.kube.update(node)
-                    await self.kube.delete("v1", "Node", ko.name_of(node))
-                except NotFoundError:
-                    pass
+        if self.nodes is not None and self.nodes.has_synced:
+            # indexed lookups (shared cache objects → copy before mutating)
+            matches: dict = {}
+            if pid:
+                for n in self.nodes.by_index("providerID", pid):
+                    matches[ko.name_of(n)] = n
+            for n in self.nodes.by_index("agentpool", name):
+                matches.setdefault(ko.name_of(n), n)
+            leaked_nodes = [ko.deep_copy(n) for n in matches.values()]
+        else:
+            leaked_nodes = [
+                node
+                for node in await self.kube.list("v1", "Node")
+                if (ko.provider_id_of(node) == pid and pid)
+                or ko.labels_of(node).get(karpv1.AGENTPOOL_LABEL_KEY) == name
+            ]
+        for node in leaked_nodes:
+            try:
+                # strip our finalizer first: the instance is already gone,
+                # there is nothing left to drain
+                if ko.remove_finalizer(node, karpv1.TERMINATION_FINALIZER):
+                    await self.kube.update(node)
+                await self.kube.delete("v1", "Node", ko.name_of(node))
+            except NotFoundError:
+                pass
 
 
 class NodeClaimGCController:
@@ -129,10 +167,18 @@ class NodeClaimGCController:
         cloud: CloudProvider,
         recorder: EventRecorder,
         interval: float = GC_INTERVAL,
+        nodes=None,  # Optional[Informer]: indexed node lookups
     ):
         self.kube = kube
         self.cloud = cloud
         self.recorder = recorder
+        self.nodes = nodes
+        if nodes is not None:
+            # providerID index (the same one lifecycle registers) — a full
+            # Node list per doomed claim is O(cluster) inside a loop
+            nodes.add_index(
+                "providerID", lambda o: o.get("spec", {}).get("providerID") or None
+            )
         self.controller = SingletonController(self.NAME, self.reconcile, interval)
 
     async def reconcile(self, key: str) -> Optional[Result]:
@@ -182,6 +228,9 @@ class NodeClaimGCController:
     async def _node_for(self, provider_id: str) -> Optional[dict]:
         if not provider_id:
             return None
+        if self.nodes is not None and self.nodes.has_synced:
+            matches = self.nodes.by_index("providerID", provider_id)
+            return matches[0] if matches else None  # read-only use
         for node in await self.kube.list("v1", "Node"):
             if ko.provider_id_of(node) == provider_id:
                 return node

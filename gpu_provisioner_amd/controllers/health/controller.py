@@ -40,12 +40,20 @@ class HealthController:
         cloud: CloudProvider,
         recorder: EventRecorder,
         nodes: Informer,
+        nodeclaims: Optional[Informer] = None,
         workers: int = 16,
     ):
         self.kube = kube
         self.cloud = cloud
         self.recorder = recorder
         self.nodes = nodes
+        self.nodeclaims = nodeclaims
+        if nodeclaims is not None:
+            # same providerID index every other controller uses — an O(cluster)
+            # NodeClaim list per unhealthy node does not belong on this path
+            nodeclaims.add_index(
+                "providerID", lambda o: o.get("status", {}).get("providerID") or None
+            )
         self.controller = Controller(self.NAME, self.reconcile, workers=workers)
         self._first_seen: dict = {}  # (node, cond_type, status) -> first observed
         nodes.add_handler(self._on_node_event)
@@ -142,9 +150,18 @@ class HealthController:
         pid = ko.provider_id_of(node)
         if not pid:
             return None
-        claims = [
-            nc
-            for nc in await self.kube.list(karpv1.API_VERSION, karpv1.KIND_NODECLAIM)
-            if karpv1.provider_id_of(nc) == pid and karpv1.is_managed(nc)
-        ]
-        return claims[0] if len(claims) == 1 else None
+        if self.nodeclaims is not None and self.nodeclaims.has_synced:
+            claims = [
+                nc
+                for nc in self.nodeclaims.by_index("providerID", pid)
+                if karpv1.is_managed(nc)
+            ]
+        else:
+            claims = [
+                nc
+                for nc in await self.kube.list(karpv1.API_VERSION, karpv1.KIND_NODECLAIM)
+                if karpv1.provider_id_of(nc) == pid and karpv1.is_managed(nc)
+            ]
+        # private copy: the caller stamps the termination-timestamp annotation
+        # in place, which must never mutate the shared informer-cache object
+        return ko.deep_copy(claims[0]) if len(claims) == 1 else None

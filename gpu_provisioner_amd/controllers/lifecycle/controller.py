@@ -150,7 +150,13 @@ class LifecycleController:
         if self.nodeclaims.has_synced:
             cached = self.nodeclaims.get(key)
             if cached is not None:
-                rv = int(ko.meta(cached).get("resourceVersion") or 0)
+                # k8s resourceVersion is an opaque string; an unparseable rv
+                # must fall through to a fresh apiserver GET, not blow up the
+                # reconcile into a hot error-retry loop
+                try:
+                    rv = int(ko.meta(cached).get("resourceVersion") or 0)
+                except (TypeError, ValueError):
+                    rv = -1
                 if rv >= self._written_rv.get(key, 0):
                     nodeclaim = ko.deep_copy(cached)
             elif key not in self._written_rv:
@@ -403,8 +409,12 @@ class LifecycleController:
             {"metadata": {"labels": {karpv1.NODE_INITIALIZED_LABEL_KEY: "true"}}},
         )
         status = nodeclaim.setdefault("status", {})
-        status["capacity"] = ko.node_capacity(node)
-        status["allocatable"] = ko.node_allocatable(node)
+        # `node` can be the SHARED informer-cache object (_nodes_by_provider_id
+        # returns it uncopied); aliasing its capacity/allocatable subtrees into
+        # NodeClaim status would let a later in-place status edit silently
+        # corrupt the cache for every other reader — copy at the boundary
+        status["capacity"] = ko.deep_copy(ko.node_capacity(node))
+        status["allocatable"] = ko.deep_copy(ko.node_allocatable(node))
         ko.set_condition(nodeclaim, karpv1.COND_INITIALIZED, ko.CONDITION_TRUE, "Initialized")
         ko.set_condition(nodeclaim, karpv1.COND_READY, ko.CONDITION_TRUE, "Ready")
         await self._patch_status(nodeclaim)

@@ -144,7 +144,11 @@ class TerminationController:
         if self.nodes.has_synced:
             cached = self.nodes.get(key)
             if cached is not None:
-                rv = int(cached.get("metadata", {}).get("resourceVersion") or 0)
+                # opaque-string rv: parse failure → fresh apiserver GET below
+                try:
+                    rv = int(cached.get("metadata", {}).get("resourceVersion") or 0)
+                except (TypeError, ValueError):
+                    rv = -1
                 if rv >= self._written_rv.get(key, 0):
                     node = ko.deep_copy(cached)
             elif key not in self._written_rv:
@@ -349,13 +353,18 @@ class TerminationController:
     async def _volume_attachments(self, node_name: str) -> list:
         if self.volumeattachments is not None and self.volumeattachments.has_synced:
             return self.volumeattachments.by_index("nodeName", node_name)
-        try:
-            return await self.kube.list(
-                "storage.k8s.io/v1", "VolumeAttachment",
-                field_selector=f"spec.nodeName={node_name}",
-            )
-        except Exception:
-            return []
+        # Fallback before informer sync: list ALL and filter client-side.
+        # A real kube-apiserver rejects field selectors on VolumeAttachment
+        # (only metadata.name/namespace are registered for it — unlike Pod's
+        # spec.nodeName) with 400; the previous `spec.nodeName=` +
+        # swallow-everything fallback would report zero pending volumes and
+        # silently skip the volume-detach wait. The reference uses an indexed
+        # cache for exactly this reason (vendor/.../operator.go:250-293).
+        return [
+            va
+            for va in await self.kube.list("storage.k8s.io/v1", "VolumeAttachment")
+            if va.get("spec", {}).get("nodeName") == node_name
+        ]
 
     def _node_termination_time(self, node: dict, nodeclaim: Optional[dict]):
         """Deadline after which drain/volume waits are cut short: the

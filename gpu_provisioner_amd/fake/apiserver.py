@@ -46,6 +46,61 @@ from ..kube.client import (
 _WATCH_HISTORY = 4096  # events kept for resourceVersion resume
 _WATCH_BROKEN = "__WATCH_BROKEN__"  # sentinel: simulated dropped stream
 
+# Field-selector support matrix of a real kube-apiserver. Every resource
+# accepts metadata.name/metadata.namespace; a handful register extra
+# selectable fields in their registry strategy (upstream
+# pkg/registry/core/pod/strategy.go ToSelectableFields et al.). Everything
+# else — including VolumeAttachment, Lease, and any CRD that doesn't declare
+# spec.selectableFields — gets "field label not supported" with HTTP 400.
+# The fake enforces the same matrix so controller code that would break on a
+# real apiserver breaks here too (a `spec.nodeName` selector on
+# VolumeAttachment once slipped through and silently skipped the
+# volume-detach wait).
+_SELECTABLE_FIELDS = {
+    ("v1", "Pod"): {
+        "spec.nodeName",
+        "spec.schedulerName",
+        "spec.restartPolicy",
+        "spec.serviceAccountName",
+        "spec.hostNetwork",
+        "status.phase",
+        "status.podIP",
+        "status.nominatedNodeName",
+    },
+    ("v1", "Node"): {"spec.unschedulable"},
+    ("v1", "Namespace"): {"status.phase", "name"},
+    ("v1", "Secret"): {"type"},
+    ("v1", "Event"): {
+        "involvedObject.kind",
+        "involvedObject.namespace",
+        "involvedObject.name",
+        "involvedObject.uid",
+        "involvedObject.apiVersion",
+        "involvedObject.resourceVersion",
+        "involvedObject.fieldPath",
+        "reason",
+        "reportingComponent",
+        "source",
+        "type",
+    },
+}
+_GENERIC_FIELDS = {"metadata.name", "metadata.namespace"}
+
+
+def validate_field_selector(api_version: str, kind: str, selector: str) -> None:
+    """Raise InvalidError (HTTP 400 shape) for selectors a real apiserver
+    would reject on this resource."""
+    if not selector:
+        return
+    allowed = _GENERIC_FIELDS | _SELECTABLE_FIELDS.get((api_version, kind), set())
+    for part in selector.split(","):
+        part = part.strip()
+        if not part:
+            continue
+        key = part.split("!=" if "!=" in part else "=", 1)[0].strip()
+        if key not in allowed:
+            raise InvalidError(f"field label not supported: {key}")
+
 
 def _key(namespace: str, name: str) -> tuple:
     return (namespace or "", name)
@@ -120,6 +175,7 @@ class InMemoryAPIServer:
     ) -> tuple:
         async with self._lock:
             gvk = _gvk(api_version, kind)
+            validate_field_selector(api_version, kind, field_selector)
             self._fire_reactors("list", gvk, None)
             sel = LabelSelector.parse(label_selector) if label_selector else None
             out = []

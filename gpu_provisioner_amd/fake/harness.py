@@ -121,10 +121,11 @@ class Harness:
         )
         self.instance_gc = InstanceGCController(
             self.kube, self.cloud, self.recorder,
-            interval=gc_interval, adoption_age=adoption_age,
+            interval=gc_interval, adoption_age=adoption_age, nodes=self.nodes,
         )
         self.nodeclaim_gc = NodeClaimGCController(
-            self.kube, self.cloud, self.recorder, interval=gc_interval
+            self.kube, self.cloud, self.recorder, interval=gc_interval,
+            nodes=self.nodes,
         )
         self.controllers += [
             self.eviction_queue, self.lifecycle, self.termination,
@@ -132,7 +133,7 @@ class Harness:
         ]
         if with_health:
             self.health = HealthController(
-                self.kube, self.cloud, self.recorder, self.nodes
+                self.kube, self.cloud, self.recorder, self.nodes, self.nodeclaims
             )
             self.controllers.append(self.health)
         if with_drift:

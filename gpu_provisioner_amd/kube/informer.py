@@ -55,9 +55,23 @@ class Informer:
 
     def add_index(self, name: str, fn: Callable) -> None:
         """fn(obj) -> Optional[str] | list[str]: index value(s) for the object.
-        Idempotent: re-registering an existing index name is a no-op."""
-        if name not in self._indexes:
-            self._indexes[name] = (fn, defaultdict(set))
+        Idempotent: re-registering an existing index name is a no-op.
+
+        Like add_handler's replay, an index registered after sync is
+        backfilled from the current cache — a late-bound index (e.g. via
+        set_nodes_informer) must not silently return empty by_index results."""
+        if name in self._indexes:
+            return
+        idx: defaultdict = defaultdict(set)
+        self._indexes[name] = (fn, idx)
+        for key, obj in self._cache.items():
+            vals = fn(obj)
+            if vals is None:
+                continue
+            if isinstance(vals, str):
+                vals = [vals]
+            for v in vals:
+                idx[v].add(key)
 
     def has_index(self, name: str) -> bool:
         return name in self._indexes

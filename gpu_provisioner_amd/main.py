@@ -66,13 +66,13 @@ def build_manager(kube, options: Options, cloud_provider, version: str = "0.1.0"
         TerminationController(
             kube, cloud, recorder, nodes, nodeclaims, pods, volumeattachments, eviction_queue
         ),
-        InstanceGCController(kube, cloud, recorder),
-        NodeClaimGCController(kube, cloud, recorder),
+        InstanceGCController(kube, cloud, recorder, nodes=nodes),
+        NodeClaimGCController(kube, cloud, recorder, nodes=nodes),
     ]
     # node.health is gated on RepairPolicies + the NodeRepair feature gate
     # (reference vendor/.../controllers/controllers.go:109-111)
     if options.feature_gates.node_repair and cloud.repair_policies():
-        controllers.append(HealthController(kube, cloud, recorder, nodes))
+        controllers.append(HealthController(kube, cloud, recorder, nodes, nodeclaims))
     # drift detection (net-new: the reference stubs IsDrifted); replacement
     # only when DriftReplace is explicitly gated on
     if options.feature_gates.drift:

@@ -284,18 +284,39 @@ def node_condition_from_report(report: NodeReport) -> dict:
 
 
 async def patch_node_condition(kube, node_name: str, report: NodeReport) -> None:
-    """Merge the AMDGPUHealthy condition into the Node's status conditions."""
+    """Merge the AMDGPUHealthy condition into the Node's status conditions.
+
+    The status write carries the read's resourceVersion as an optimistic
+    lock and retries on conflict: a merge patch of the full conditions list
+    from an unconditioned GET races the kubelet's concurrent status writes
+    and can clobber or resurrect kubelet-owned conditions (the same
+    lost-update class TerminationController._set_nodeclaim_condition guards
+    against)."""
     from .kube import objects as ko
+    from .kube.client import ConflictError
 
     cond = node_condition_from_report(report)
-    node = await kube.get("v1", "Node", node_name)
-    ko.set_condition(node, cond["type"], cond["status"], cond["reason"], cond["message"])
-    await kube.patch(
-        "v1",
-        "Node",
-        node_name,
-        {"status": {"conditions": node["status"]["conditions"]}},
-        subresource="status",
+    for _ in range(5):
+        node = await kube.get("v1", "Node", node_name)
+        ko.set_condition(node, cond["type"], cond["status"], cond["reason"], cond["message"])
+        try:
+            await kube.patch(
+                "v1",
+                "Node",
+                node_name,
+                {
+                    "metadata": {
+                        "resourceVersion": node.get("metadata", {}).get("resourceVersion")
+                    },
+                    "status": {"conditions": node["status"]["conditions"]},
+                },
+                subresource="status",
+            )
+            return
+        except ConflictError:
+            continue
+    raise NodeAgentError(
+        f"node {node_name}: AMDGPUHealthy condition patch abandoned after repeated conflicts"
     )
 
 

@@ -140,9 +140,15 @@ class InstanceTypeProvider:
         it = self._types.get(vm_size)
         if it is not None:
             return karpv1.AMD_GPU_RESOURCE in it.capacity
-        # Conservative fallback for SKUs outside the catalog: Azure AMD
-        # Instinct families carry the accelerator model in the name.
-        return "_MI3" in vm_size or vm_size.startswith("Standard_ND")
+        # Conservative fallback for SKUs outside the catalog: only Azure AMD
+        # Instinct families, which carry the accelerator model in the name
+        # (…_MI300X_…, …_MI355X_…). A bare Standard_ND*/Standard_NC* prefix
+        # must NOT match — Azure's ND/NC families are mostly NVIDIA (e.g.
+        # ND H100 v5, ND A100 v4) and stamping a ROCm gpuProfile on those
+        # pools would break them.
+        import re
+
+        return re.search(r"_MI\d{3}", vm_size) is not None
 
     def gpu_count(self, vm_size: str) -> int:
         it = self._types.get(vm_size)

@@ -226,3 +226,28 @@ def test_wait_until_registries_do_not_leak():
         await inf.stop()
 
     run(main())
+
+
+def test_add_index_after_sync_backfills_existing_cache():
+    """A late-registered index (the set_nodes_informer path binds indexes
+    after informers may already be synced) must be backfilled from the
+    current cache, mirroring add_handler's replay — otherwise by_index
+    silently returns empty for pre-existing objects."""
+
+    async def main():
+        server = InMemoryAPIServer()
+        kube = InMemoryClient(server)
+        await kube.create(mk_node("pre", "azure:///pid-pre"))
+        inf = Informer(kube, "v1", "Node")
+        inf.start()
+        await asyncio.wait_for(inf.wait_for_sync(), 5)
+        # register AFTER sync — "pre" is already cached
+        inf.add_index("providerID", lambda o: o.get("spec", {}).get("providerID") or None)
+        got = inf.by_index("providerID", "azure:///pid-pre")
+        assert [o["metadata"]["name"] for o in got] == ["pre"]
+        # idempotent re-registration stays a no-op
+        inf.add_index("providerID", lambda o: None)
+        assert inf.by_index("providerID", "azure:///pid-pre")
+        await inf.stop()
+
+    run(main())

@@ -120,6 +120,18 @@ def test_new_agent_pool_object_cpu_sku_has_no_gpu_profile():
     assert karpv1.AMD_GPU_COUNT_LABEL_KEY not in pool["properties"]["nodeLabels"]
 
 
+def test_new_agent_pool_object_nvidia_sku_gets_no_rocm_profile():
+    """Out-of-catalog NVIDIA ND/NC SKUs must never receive the ROCm
+    bootstrap (gpuProfile/kubeletConfig/linuxOSConfig) — VERDICT r01 #4."""
+    provider, *_ = make_provider()
+    for sku in ("Standard_ND96asr_v4", "Standard_ND96isr_H100_v5", "Standard_NC24ads_A100_v4"):
+        pool = provider.new_agent_pool_object(nodeclaim(vm=sku), sku)
+        props = pool["properties"]
+        assert "gpuProfile" not in props, sku
+        assert "kubeletConfig" not in props, sku
+        assert karpv1.AMD_GPU_COUNT_LABEL_KEY not in props["nodeLabels"], sku
+
+
 @pytest.mark.parametrize(
     "annotation,expected",
     [

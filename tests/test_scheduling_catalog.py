@@ -128,6 +128,13 @@ def test_catalog_gpu_detection():
     assert cat.gpu_count("Standard_D4s_v5") == 0
     # AMD-family fallback for SKUs outside the catalog
     assert cat.is_gpu_sku("Standard_ND256is_MI455X_v7")
+    # NVIDIA ND/NC SKUs outside the catalog must NOT be treated as AMD GPU
+    # SKUs — a bare Standard_ND prefix match would stamp a ROCm gpuProfile
+    # on an ND A100/H100 pool
+    assert not cat.is_gpu_sku("Standard_ND96asr_v4")  # NVIDIA A100
+    assert not cat.is_gpu_sku("Standard_ND96isr_H100_v5")  # NVIDIA H100
+    assert not cat.is_gpu_sku("Standard_NC24ads_A100_v4")
+    assert cat.gpu_count("Standard_ND96asr_v4") == 0
 
 
 def test_mi300x_previous_generation_in_catalog():
