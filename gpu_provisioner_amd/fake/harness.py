@@ -37,7 +37,9 @@ class Harness:
         region: str = "eastus2",
         repair_toleration: Optional[float] = None,
         gpu_repair_toleration: Optional[float] = None,
+        gc_pacer: bool = True,
     ):
+        self._gc_pacer_enabled = gc_pacer
         self.server = InMemoryAPIServer()
         self.kube = InMemoryClient(self.server)
         self.catalog = InstanceTypeProvider(region)
@@ -168,9 +170,19 @@ class Harness:
         await self.informers.wait_for_sync()
         for c in self.controllers:
             c.controller.start()
+        if self._gc_pacer_enabled:
+            # production Manager topology: paced cyclic GC once steady
+            from ..operator.gcpacer import GCPacer
+
+            self.gc_pacer = GCPacer()
+            self.gc_pacer.engage()
+        else:
+            self.gc_pacer = None
         self._started = True
 
     async def stop(self) -> None:
+        if getattr(self, "gc_pacer", None) is not None:
+            await self.gc_pacer.disengage()
         for c in self.controllers:
             await c.controller.stop()
         await self.informers.stop_all()

@@ -165,6 +165,16 @@ ARM_RETRIES = _get_or_create(
     ("operation",),
 )
 
+# -- runtime ------------------------------------------------------------------
+
+GC_PAUSE_SECONDS = _get_or_create(
+    Histogram,
+    "gpu_provisioner_gc_pause_seconds",
+    "Duration of paced CPython garbage-collection passes (operator.gcpacer)",
+    ("generation",),
+    buckets=(0.001, 0.0025, 0.005, 0.01, 0.025, 0.05, 0.1, 0.25),
+)
+
 # -- build info ---------------------------------------------------------------
 
 BUILD_INFO = _get_or_create(

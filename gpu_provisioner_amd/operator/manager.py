@@ -23,6 +23,7 @@ import uvicorn
 from ..kube.client import KubeClient, NotFoundError
 from ..kube.informer import InformerFactory
 from ..metrics.registry import BUILD_INFO
+from .gcpacer import GCPacer
 from .leaderelection import LeaderElector
 from .options import Options
 
@@ -48,6 +49,10 @@ class Manager:
         self.lease_duration = lease_duration
         self.renew_interval = renew_interval
         self.controllers: list = []
+        # bounded-tail cyclic GC (see operator/gcpacer.py); opt out via options
+        self.gc_pacer: Optional[GCPacer] = (
+            GCPacer() if getattr(options, "gc_pacer", True) else None
+        )
         self._elector: Optional[LeaderElector] = None
         self._server_tasks: list = []
         self._elector_task: Optional[asyncio.Task] = None
@@ -145,6 +150,10 @@ class Manager:
             )
         else:
             await self._start_controllers()
+        if self.gc_pacer is not None:
+            # engage at steady state: caches synced, controllers built —
+            # that heap is what gets frozen out of future full collections
+            self.gc_pacer.engage()
         self._started.set()
 
     async def _start_controllers(self) -> None:
@@ -176,6 +185,8 @@ class Manager:
         await self.stop()
 
     async def stop(self) -> None:
+        if self.gc_pacer is not None:
+            await self.gc_pacer.disengage()
         if self._elector is not None:
             await self._elector.release()
         if self._elector_task is not None:

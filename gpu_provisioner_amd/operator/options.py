@@ -62,6 +62,9 @@ class Options:
     leader_election_namespace: str = "kube-system"
     disable_webhook: bool = True
     log_level: str = "info"
+    # paced cyclic GC (operator/gcpacer.py) — on by default; GC_PACER=false
+    # restores CPython's automatic full collections
+    gc_pacer: bool = True
     feature_gates: FeatureGates = field(default_factory=FeatureGates)
 
     @classmethod
@@ -107,6 +110,11 @@ class Options:
         )
         p.add_argument("--log-level", default=envv("LOG_LEVEL", "info"))
         p.add_argument(
+            "--gc-pacer",
+            default=envv("GC_PACER", "true").lower() == "true",
+            action="store_true",
+        )
+        p.add_argument(
             "--feature-gates", default=envv("FEATURE_GATES", "NodeRepair=true")
         )
         args = p.parse_args(argv if argv is not None else [])
@@ -121,5 +129,6 @@ class Options:
             leader_election_namespace=args.leader_election_namespace,
             disable_webhook=args.disable_webhook,
             log_level=args.log_level,
+            gc_pacer=args.gc_pacer,
             feature_gates=FeatureGates.parse(args.feature_gates),
         )
