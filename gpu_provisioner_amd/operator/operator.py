@@ -44,12 +44,18 @@ class Operator:
         extra_headers = {}
         if env.get("E2E_TEST_MODE", "").lower() == "true":
             extra_headers["X-Kaito-E2E"] = env.get("E2E_SCENARIO", "true")
+        # pinned ARM api-version + request schema travel together
+        # (armschema.py; ARM_API_PROFILE env selects stable/gpu-preview)
+        from ..providers.instance.armschema import profile_from_env
+
+        arm_profile = profile_from_env(environ)
         self.agent_pools = ARMAgentPoolsClient(
             self.credential,
             self.config.subscription_id,
             endpoint=self.config.arm_endpoint,
             user_agent=self.config.user_agent,
             extra_headers=extra_headers,
+            api_version=arm_profile.api_version,
         )
         self.catalog = InstanceTypeProvider(region=self.config.location)
         self.instances = InstanceProvider(
@@ -58,6 +64,7 @@ class Operator:
             self.catalog,
             resource_group=self.config.resource_group,
             cluster_name=self.config.cluster_name,
+            arm_profile=arm_profile,
         )
         self.cloud_provider = AzureCloudProvider(self.instances, self.catalog)
         log.info(

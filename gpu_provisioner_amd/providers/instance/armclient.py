@@ -20,10 +20,14 @@ import httpx
 from ...auth.cred import TokenCredential
 from ...metrics.registry import ARM_REQUEST_DURATION, ARM_RETRIES
 from .armapi import AgentPoolsAPI, ARMError, LROPoller
+from .armschema import PROFILE_STABLE
 
 log = logging.getLogger(__name__)
 
-API_VERSION = "2024-09-01"
+# default pinned api-version; overridden per-client by the ArmApiProfile
+# (armschema.py) so the request schema and the api-version always travel
+# together
+API_VERSION = PROFILE_STABLE.api_version
 MAX_RETRIES = 20  # reference armopts.go:34-40
 RETRY_BASE_SECONDS = 5.0
 RETRY_CAP_SECONDS = 60.0
@@ -93,9 +97,11 @@ class ARMAgentPoolsClient(AgentPoolsAPI):
         lro_poll_interval: float = 5.0,
         max_retries: int = MAX_RETRIES,
         extra_headers: Optional[dict] = None,
+        api_version: str = API_VERSION,
     ):
         self.credential = credential
         self.subscription_id = subscription_id
+        self.api_version = api_version
         self.endpoint = endpoint.rstrip("/")
         self.lro_poll_interval = lro_poll_interval
         self.max_retries = max_retries
@@ -130,7 +136,7 @@ class ARMAgentPoolsClient(AgentPoolsAPI):
             try:
                 # nextLink/Azure-AsyncOperation URLs already carry their query
                 # string; httpx `params` would replace it
-                params = None if "?" in url else {"api-version": API_VERSION}
+                params = None if "?" in url else {"api-version": self.api_version}
                 resp = await self.http.request(
                     method,
                     url,

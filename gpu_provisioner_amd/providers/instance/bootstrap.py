@@ -20,6 +20,8 @@ bootstrap here is explicit:
 """
 from __future__ import annotations
 
+from typing import Optional
+
 from ...apis import v1 as karpv1
 from ..instancetype.catalog import (
     GFX_ARCH,
@@ -29,10 +31,16 @@ from ..instancetype.catalog import (
     XGMI_GBPS_PER_LINK,
     XGMI_LINKS_PER_GPU,
 )
+from .armschema import ArmApiProfile, PROFILE_STABLE
 
 OSSKU_UBUNTU = "Ubuntu"
 OSSKU_AZURELINUX = "AzureLinux"
 
+# ROCm stack versions the preview gpuProfile requests. EXTRAPOLATED for the
+# MI355X generation (no published swagger pins these — see armschema.py);
+# only emitted under the explicitly-selected gpu-preview ARM profile. On the
+# stable profile the same stack is installed by the chart's amdgpu-driver +
+# device-plugin DaemonSets.
 ROCM_VERSION = "6.4"
 AMDGPU_DRIVER_VERSION = "6.10.5"
 DEVICE_PLUGIN_IMAGE = "rocm/k8s-device-plugin:latest"
@@ -52,15 +60,23 @@ def determine_os_sku(image_family_annotation: str) -> str:
     return _IMAGE_FAMILY_TO_OSSKU.get((image_family_annotation or "").strip().lower(), OSSKU_UBUNTU)
 
 
-def rocm_gpu_profile() -> dict:
+def rocm_gpu_profile(profile: ArmApiProfile = PROFILE_STABLE) -> Optional[dict]:
     """AgentPool gpuProfile requesting AMD driver installation (the AKS-side
-    switch that provisions amdgpu instead of the NVIDIA stack)."""
-    return {
+    switch that provisions amdgpu instead of the NVIDIA stack).
+
+    Returns None when the pinned api-version defines no gpuProfile (the
+    stable default) — the builder must then omit the property entirely and
+    driver installation falls to the chart's DaemonSets. Under gpu-preview
+    only the profile's declared sub-fields are emitted."""
+    if not profile.gpu_profile_fields:
+        return None
+    full = {
         "driver": "Install",
         "driverType": "ROCm",
         "driverVersion": AMDGPU_DRIVER_VERSION,
         "rocmVersion": ROCM_VERSION,
     }
+    return {k: v for k, v in full.items() if k in profile.gpu_profile_fields}
 
 
 def rocm_kubelet_config() -> dict:

@@ -81,7 +81,10 @@ class InstanceProvider:
         *,
         node_wait_attempts: int = NODE_WAIT_ATTEMPTS,
         node_wait_interval: float = NODE_WAIT_INTERVAL,
+        arm_profile=None,  # armschema.ArmApiProfile; None → ARM_API_PROFILE env / stable
     ):
+        from .armschema import profile_from_env
+
         self.agent_pools = agent_pools
         self.kube = kube
         self.catalog = catalog
@@ -89,6 +92,7 @@ class InstanceProvider:
         self.cluster_name = cluster_name
         self.node_wait_attempts = node_wait_attempts
         self.node_wait_interval = node_wait_interval
+        self.arm_profile = arm_profile if arm_profile is not None else profile_from_env()
         # optional Node informer for cache-backed lookups (the reference
         # reads through controller-runtime's cached client; an apiserver
         # LIST per wait attempt is O(cluster) and shows at 128 concurrent)
@@ -121,7 +125,12 @@ class InstanceProvider:
                 condition_reason="InvalidName",
             )
         vm_size = self._pick_vm_size(nodeclaim)
-        pool = self.new_agent_pool_object(nodeclaim, vm_size)
+        from .armschema import SchemaViolation
+
+        try:
+            pool = self.new_agent_pool_object(nodeclaim, vm_size)
+        except SchemaViolation as e:
+            raise CreateError(str(e), condition_reason="InvalidRequest") from e
         try:
             poller = await self.agent_pools.begin_create_or_update(
                 self.resource_group, self.cluster_name, name, pool
@@ -249,10 +258,19 @@ class InstanceProvider:
             props["osDiskSizeGB"] = max(1, int(ko.qty(disk).value / 2**30))
         if self.catalog.is_gpu_sku(vm_size):
             gpus = self.catalog.gpu_count(vm_size)
-            props["gpuProfile"] = bootstrap.rocm_gpu_profile()
+            gpu_profile = bootstrap.rocm_gpu_profile(self.arm_profile)
+            if gpu_profile is not None:
+                props["gpuProfile"] = gpu_profile
             props["kubeletConfig"] = bootstrap.rocm_kubelet_config()
             props["linuxOSConfig"] = bootstrap.rocm_linux_os_config(gpus)
-        return {"name": ko.name_of(nodeclaim), "properties": props}
+        pool = {"name": ko.name_of(nodeclaim), "properties": props}
+        # contract check against the pinned api-version (armschema.py): a
+        # field outside the schema must fail HERE, in our own tests, not be
+        # silently dropped by ARM in production
+        from .armschema import validate_agent_pool
+
+        validate_agent_pool(pool, self.arm_profile)
+        return pool
 
     async def _wait_for_node(self, pool: str) -> tuple:
         """Wait for the Node object + providerID (reference instance.go:123-149,

@@ -23,10 +23,15 @@ HBM_PER_GPU_GB = 288
 XGMI_LINKS_PER_GPU = 7
 XGMI_GBPS_PER_LINK = 153
 
-# SKU table. Azure's AMD Instinct family uses the NDis-MI naming scheme
-# (cf. Standard_ND96isr_MI300X_v5 for the MI300X generation); the MI355X
-# generation is modeled the same way. Fields: vCPU, memory GiB, GPU count,
-# ephemeral OS disk ceiling GiB, price-per-hour (on-demand, list), zones.
+# SKU table — PROVENANCE: Azure's AMD Instinct family uses the NDis-MI
+# naming scheme (Standard_ND96isr_MI300X_v5 is the published MI300X SKU);
+# the MI355X-generation names/sizes below are an ACKNOWLEDGED EXTRAPOLATION
+# of that scheme (no MI355X SKU list is publishable at build time, and this
+# environment is offline). Deployments override or extend the table without
+# a code change via GPU_PROV_SKU_FILE (YAML/JSON list of SKU dicts, same
+# fields) — see InstanceTypeProvider.__init__.
+# Fields: vCPU, memory GiB, GPU count, ephemeral OS disk ceiling GiB,
+# price-per-hour (on-demand, list), zones.
 _SKUS = [
     {
         "name": "Standard_ND128isr_MI355X_v6",
@@ -123,9 +128,40 @@ def _build(sku: dict, region: str, zones=_DEFAULT_ZONES) -> InstanceType:
 class InstanceTypeProvider:
     """Catalog provider: list SKUs, resolve one by name, answer GPU questions."""
 
-    def __init__(self, region: str = "eastus2"):
+    def __init__(self, region: str = "eastus2", sku_file: Optional[str] = None):
+        import os
+
         self.region = region
-        self._types = {s["name"]: _build(s, region) for s in _SKUS}
+        skus = list(_SKUS)
+        path = sku_file if sku_file is not None else os.environ.get("GPU_PROV_SKU_FILE", "")
+        if path:
+            skus = self._merge_sku_file(skus, path)
+        self._types = {s["name"]: _build(s, region) for s in skus}
+
+    @staticmethod
+    def _merge_sku_file(skus: list, path: str) -> list:
+        """Load a YAML/JSON SKU list; entries override built-ins by name.
+        An entry of just {name: ..., remove: true} drops a built-in SKU."""
+        import yaml
+
+        with open(path) as f:
+            loaded = yaml.safe_load(f) or []
+        if not isinstance(loaded, list):
+            raise ValueError(f"{path}: expected a list of SKU entries")
+        by_name = {s["name"]: s for s in skus}
+        for entry in loaded:
+            name = entry.get("name")
+            if not name:
+                raise ValueError(f"{path}: SKU entry missing 'name': {entry}")
+            if entry.get("remove"):
+                by_name.pop(name, None)
+                continue
+            required = {"vcpu", "memory_gib", "gpus", "max_os_disk_gib", "price"}
+            missing = required - set(entry)
+            if missing:
+                raise ValueError(f"{path}: SKU {name} missing fields {sorted(missing)}")
+            by_name[name] = entry
+        return list(by_name.values())
 
     def list(self) -> list:
         return list(self._types.values())

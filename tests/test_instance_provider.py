@@ -108,7 +108,10 @@ def test_new_agent_pool_object_shape():
     assert labels[karpv1.AMD_GPU_COUNT_LABEL_KEY] == "8"
     assert labels[karpv1.AMD_GPU_VRAM_LABEL_KEY] == "288G"
     assert labels[karpv1.XGMI_TOPOLOGY_LABEL_KEY].startswith("8x-7l-")
-    assert props["gpuProfile"]["driverType"] == "ROCm"
+    # stable ARM profile (the default): no gpuProfile property — it is not
+    # in the pinned 2024-09-01 agentPools schema (armschema.py); the ROCm
+    # stack comes from the chart's DaemonSets
+    assert "gpuProfile" not in props
     assert props["kubeletConfig"]["topologyManagerPolicy"] == "single-numa-node"
 
 

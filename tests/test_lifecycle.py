@@ -59,7 +59,10 @@ def test_provision_to_initialized_full_path():
             props = pool["properties"]
             assert props["count"] == 1
             assert props["vmSize"] == "Standard_ND128isr_MI355X_v6"
-            assert props["gpuProfile"]["driverType"] == "ROCm"
+            # stable ARM profile: ROCm bootstrap via kubeletConfig/labels,
+            # no gpuProfile (not in the pinned 2024-09-01 schema — armschema.py)
+            assert "gpuProfile" not in props
+            assert props["kubeletConfig"]["topologyManagerPolicy"] == "single-numa-node"
         finally:
             await h.stop()
 

@@ -204,3 +204,31 @@ def test_event_recorder_dedupes():
         assert reasons == ["Launched", "Registered"]  # dedupe collapsed 5→1
 
     run(main())
+
+
+def test_catalog_sku_file_override(tmp_path):
+    """GPU_PROV_SKU_FILE-style override: add a SKU, replace a built-in,
+    remove a built-in — deployments track real SKU lists without code
+    changes (the built-in MI355X names are acknowledged extrapolations)."""
+    import yaml
+
+    f = tmp_path / "skus.yaml"
+    f.write_text(yaml.safe_dump([
+        {"name": "Standard_ND96isr_MI400X_v7", "vcpu": 96, "memory_gib": 2300,
+         "gpus": 8, "max_os_disk_gib": 4096, "price": 90.0,
+         "vram_gb": 432, "arch": "gfx1000x"},
+        {"name": "Standard_ND128isr_MI355X_v6", "vcpu": 128, "memory_gib": 2048,
+         "gpus": 8, "max_os_disk_gib": 4096, "price": 60.0},  # price corrected
+        {"name": "Standard_ND32is_MI355X_v6", "remove": True},
+    ]))
+    cat = InstanceTypeProvider(sku_file=str(f))
+    assert cat.get("Standard_ND96isr_MI400X_v7") is not None
+    assert cat.gpu_count("Standard_ND96isr_MI400X_v7") == 8
+    assert cat.get("Standard_ND32is_MI355X_v6") is None
+    flagship = cat.get("Standard_ND128isr_MI355X_v6")
+    assert min(o.price for o in flagship.offerings if o.capacity_type == "on-demand") == 60.0
+    # malformed entries fail loudly
+    bad = tmp_path / "bad.yaml"
+    bad.write_text(yaml.safe_dump([{"name": "X", "vcpu": 1}]))
+    with pytest.raises(ValueError, match="missing fields"):
+        InstanceTypeProvider(sku_file=str(bad))
