@@ -109,3 +109,62 @@ def test_validator_rejects_malformed():
     bad = karpv1.new_nodeclaim("bad1")
     bad["spec"] = {"requirements": [{"key": 42, "operator": "In", "values": "notalist"}]}
     assert validate(bad, schema), "validator failed to flag malformed requirements"
+
+
+# ------------------------------------------------- reference parity (r02)
+
+
+def test_crd_parity_with_reference():
+    """Field-by-field structural diff against the reference chart's CRD
+    (hack/crd_parity.py). Runs wherever the reference tree
+    is available; deliberate deltas must be documented in the script."""
+    import importlib.util
+
+    ref = "/root/reference/charts/gpu-provisioner/crds/karpenter.sh_nodeclaims.yaml"
+    if not os.path.exists(ref):
+        pytest.skip("reference tree not available")
+    spec = importlib.util.spec_from_file_location(
+        "crd_parity", os.path.join(ROOT, "hack", "crd_parity.py")
+    )
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    result = mod.compare(ref, CRD_PATH)
+    assert not result["missing"], f"reference features missing: {result['missing']}"
+    assert not result["different"], f"feature values differ: {result['different']}"
+    assert not result["columns_missing"], result["columns_missing"]
+    assert result["subresources_ok"]
+
+
+def test_crd_feature_manifest():
+    """Key validation features asserted WITHOUT the reference tree (CI
+    guard): extracted facts from the karpenter.sh/v1 schema that must not
+    regress."""
+    schema = load_schema()
+    spec = schema["properties"]["spec"]
+    assert sorted(spec["required"]) == ["nodeClassRef", "requirements"]
+    reqs = spec["properties"]["requirements"]
+    assert reqs["maxItems"] == 100
+    assert len(reqs["x-kubernetes-validations"]) == 3
+    item = reqs["items"]["properties"]
+    assert item["operator"]["enum"] == ["In", "NotIn", "Exists", "DoesNotExist", "Gt", "Lt"]
+    assert item["key"]["maxLength"] == 316
+    assert len(item["key"]["x-kubernetes-validations"]) == 4  # restricted domains
+    assert item["values"]["items"]["maxLength"] == 63
+    # quantity pattern on resource maps
+    qty_pat = spec["properties"]["resources"]["properties"]["requests"][
+        "additionalProperties"]["pattern"]
+    assert "[KMGTPE]i" in qty_pat
+    status = schema["properties"]["status"]["properties"]
+    assert status["capacity"]["additionalProperties"]["pattern"] == qty_pat
+    cond = status["conditions"]["items"]
+    assert sorted(cond["required"]) == ["lastTransitionTime", "status", "type"]
+    assert cond["properties"]["type"]["maxLength"] == 316
+    assert cond["properties"]["reason"]["pattern"].startswith("^([A-Za-z]")
+    # printer columns incl. the zone + priority-1 set
+    crd = next(yaml.safe_load_all(open(CRD_PATH)))
+    cols = {c["name"]: c for c in crd["spec"]["versions"][0]["additionalPrinterColumns"]}
+    assert {"Type", "Capacity", "Zone", "Node", "Ready", "Age",
+            "ImageID", "ID", "NodePool", "NodeClass", "Drifted"} <= set(cols)
+    assert cols["Drifted"]["priority"] == 1
+    assert "expireAfter" in spec["properties"]
+    assert spec["properties"]["expireAfter"]["default"] == "720h"
