@@ -78,6 +78,9 @@ class HTTPClient(KubeClient):
         headers = {"User-Agent": user_agent}
         if token:
             headers["Authorization"] = f"Bearer {token}"
+        if isinstance(verify, str):
+            # CA file path → ssl context (httpx deprecated verify=<str>)
+            verify = ssl.create_default_context(cafile=verify)
         # connection pool sized like the reference's armbalancer-tuned
         # transport (pkg/utils/opts/init_http_client.go:29-52)
         self.http = httpx.AsyncClient(
@@ -101,6 +104,86 @@ class HTTPClient(KubeClient):
         ca = os.path.join(SA_DIR, "ca.crt")
         verify: "str | bool" = ca if os.path.exists(ca) else True
         return cls(f"https://{host}:{port}", token=token, verify=verify, qps=qps, burst=burst)
+
+    @classmethod
+    def from_kubeconfig(
+        cls,
+        path: str = "",
+        context: str = "",
+        qps: float = 200.0,
+        burst: int = 300,
+    ) -> "HTTPClient":
+        """Client from a kubeconfig file (the live-cluster e2e path; the
+        reference harness wraps a kubeconfig the same way —
+        test/e2e/pkg/environment/common/environment.go:56-84). Supports
+        bearer tokens (inline or tokenFile), client certificates (inline
+        base64 *-data or file paths) and CA pinning; exec credential
+        plugins are rejected with a clear error."""
+        import base64
+        import tempfile
+
+        import yaml
+
+        path = path or os.environ.get("KUBECONFIG", os.path.expanduser("~/.kube/config"))
+        with open(path) as f:
+            cfg = yaml.safe_load(f)
+
+        def by_name(section: str, name: str) -> dict:
+            for entry in cfg.get(section, []):
+                if entry.get("name") == name:
+                    return entry
+            raise ValueError(f"kubeconfig {path}: no {section} entry named {name!r}")
+
+        ctx_name = context or cfg.get("current-context", "")
+        if not ctx_name:
+            raise ValueError(f"kubeconfig {path}: no current-context")
+        ctx = by_name("contexts", ctx_name)["context"]
+        cluster = by_name("clusters", ctx["cluster"])["cluster"]
+        user = by_name("users", ctx["user"])["user"]
+        if "exec" in user:
+            raise ValueError(
+                "kubeconfig uses an exec credential plugin; provide a token "
+                "or client certificate for the e2e environment"
+            )
+
+        def materialize(data_key: str, file_key: str) -> str:
+            if user.get(file_key):
+                return user[file_key]
+            if user.get(data_key):
+                tmp = tempfile.NamedTemporaryFile(delete=False, suffix=".pem")
+                tmp.write(base64.b64decode(user[data_key]))
+                tmp.close()
+                return tmp.name
+            return ""
+
+        token = user.get("token", "")
+        if not token and user.get("tokenFile"):
+            with open(user["tokenFile"]) as f:
+                token = f.read().strip()
+        verify: "ssl.SSLContext | str | bool" = True
+        if cluster.get("insecure-skip-tls-verify"):
+            verify = False
+        elif cluster.get("certificate-authority"):
+            verify = cluster["certificate-authority"]
+        elif cluster.get("certificate-authority-data"):
+            ca_tmp = tempfile.NamedTemporaryFile(delete=False, suffix=".crt")
+            ca_tmp.write(base64.b64decode(cluster["certificate-authority-data"]))
+            ca_tmp.close()
+            verify = ca_tmp.name
+        cert_file = materialize("client-certificate-data", "client-certificate")
+        key_file = materialize("client-key-data", "client-key")
+        if cert_file and key_file:
+            sslctx = ssl.create_default_context(
+                cafile=verify if isinstance(verify, str) else None
+            )
+            if verify is False:
+                sslctx.check_hostname = False
+                sslctx.verify_mode = ssl.CERT_NONE
+            sslctx.load_cert_chain(cert_file, key_file)
+            verify = sslctx
+        return cls(
+            cluster["server"], token=token, verify=verify, qps=qps, burst=burst
+        )
 
     # ------------------------------------------------------------ plumbing
 
@@ -306,6 +389,31 @@ class HTTPClient(KubeClient):
                 obj.setdefault("apiVersion", api_version)
                 obj.setdefault("kind", kind)
                 yield etype, obj
+
+    async def read_pod_log(
+        self,
+        name: str,
+        namespace: str,
+        container: str = "",
+        tail_lines: int = 0,
+        previous: bool = False,
+    ) -> str:
+        """GET /api/v1/.../pods/<name>/log — controller-log dumping for the
+        e2e harness (reference expectation.go:375 dumps controller logs on
+        spec failure)."""
+        await self._throttle()
+        params: dict = {}
+        if container:
+            params["container"] = container
+        if tail_lines:
+            params["tailLines"] = str(tail_lines)
+        if previous:
+            params["previous"] = "true"
+        resp = await self.http.get(
+            self._path("v1", "Pod", name, namespace) + "/log", params=params
+        )
+        self._raise_for(resp)
+        return resp.text
 
     async def evict(self, pod: dict, grace_period_seconds: Optional[int] = None) -> None:
         await self._throttle()

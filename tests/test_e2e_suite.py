@@ -1,9 +1,20 @@
-"""The e2e suite: the reference's 8 ginkgo scenarios
-(test/e2e/suites/suite_test.go) run against the full in-process controller
-topology + AKS simulator. Each spec drives only the public surface
-(NodeClaim/Node objects) and asserts cluster-visible outcomes, like the
-reference's Environment/Eventually harness."""
+"""The e2e suite — the reference's 8 ginkgo scenarios
+(test/e2e/suites/suite_test.go) plus scale/fault tiers.
+
+The 8 reference scenarios (spec1-spec8) are DUAL-BACKEND: each spec drives
+only the tests/e2e_env.py environment surface, so the identical functions
+run against the in-process controller topology + AKS simulator (default) or
+a LIVE cluster (`E2E_LIVE=1` + KUBECONFIG, controller deployed via the
+chart). Cloud-internal assertions (agent-pool bodies, create counters) are
+checked where the backend can observe them and skipped on live, exactly as
+the reference's ginkgo suite judges outcomes by the kube-visible surface.
+
+The scale/perf/fault specs (spec9+) intentionally stay in-process only:
+they script simulator latency, fault reactors and zero-latency churn that a
+live cluster cannot reproduce deterministically.
+"""
 import asyncio
+import os
 
 import pytest
 
@@ -12,205 +23,199 @@ from gpu_provisioner_amd.apis import v1alpha1
 from gpu_provisioner_amd.fake.harness import Harness
 from gpu_provisioner_amd.kube import objects as ko
 from tests.conftest import run
+from tests.e2e_env import make_env, spec_nodeclaim
+
+LIVE = os.environ.get("E2E_LIVE", "") == "1"
 
 
-def env(**kw) -> Harness:
-    return Harness(
-        ready_latency=0.05, plugin_latency=0.05, **kw
-    ).add_all_controllers(gc_interval=60.0)
+def run_spec(spec, timeout: float = 120.0, **env_kw):
+    """Run one dual-backend spec body against the selected environment."""
+
+    async def main():
+        env = make_env(**env_kw)
+        await env.start()
+        try:
+            await spec(env)
+        finally:
+            await env.stop()
+
+    # live expectations run on the reference's 10-min SLO; give the spec
+    # room for several of them plus cleanup
+    run(main(), timeout=3600.0 if LIVE else timeout)
 
 
-def spec_nodeclaim(name, labels, vm="Standard_ND128isr_MI355X_v6", node_class=None, annotations=None):
-    nc = karpv1.new_nodeclaim(name, labels=labels)
-    nc["spec"] = {
-        "requirements": [
-            {"key": karpv1.INSTANCE_TYPE_LABEL_KEY, "operator": "In", "values": [vm]}
-        ],
-        "resources": {"requests": {karpv1.AMD_GPU_RESOURCE: "8"}},
-        "nodeClassRef": node_class
-        or {"group": "kaito.sh", "kind": "KaitoNodeClass", "name": "default"},
-    }
-    if annotations:
-        nc["metadata"]["annotations"] = annotations
-    return nc
+# --------------------------------------------------------------- spec 1-8
+# (reference test/e2e/suites/suite_test.go — same scenarios, same order)
 
 
 def test_spec1_provision_via_workspace_label():
-    async def main():
-        h = env()
-        await h.start()
-        try:
-            await h.kube.create(spec_nodeclaim("ws1", {karpv1.KAITO_WORKSPACE_LABEL_KEY: "llm"}))
-            nc = await h.wait_initialized("ws1")
-            node = await h.kube.get("v1", "Node", nc["status"]["nodeName"])
-            assert ko.node_is_ready(node)
-            assert ko.node_allocatable(node)[karpv1.AMD_GPU_RESOURCE] == "8"
-            assert ko.labels_of(node)[karpv1.KAITO_WORKSPACE_LABEL_KEY] == "llm"
-        finally:
-            await h.stop()
+    async def spec(env):
+        await env.create(env.nodeclaim("ws1", {karpv1.KAITO_WORKSPACE_LABEL_KEY: "llm"}))
+        nc = await env.wait_initialized("ws1")
+        node = await env.kube.get("v1", "Node", nc["status"]["nodeName"])
+        assert ko.node_is_ready(node)
+        assert ko.qty(ko.node_allocatable(node)[karpv1.AMD_GPU_RESOURCE]) == ko.qty("8")
+        assert ko.labels_of(node)[karpv1.KAITO_WORKSPACE_LABEL_KEY] == "llm"
 
-    run(main())
+    run_spec(spec)
 
 
 def test_spec2_provision_via_ragengine_label():
-    async def main():
-        h = env()
-        await h.start()
-        try:
-            await h.kube.create(spec_nodeclaim("rag1", {karpv1.KAITO_RAGENGINE_LABEL_KEY: "rag"}))
-            nc = await h.wait_initialized("rag1")
-            assert karpv1.is_initialized(nc)
-        finally:
-            await h.stop()
+    async def spec(env):
+        await env.create(env.nodeclaim("rag1", {karpv1.KAITO_RAGENGINE_LABEL_KEY: "rag"}))
+        nc = await env.wait_initialized("rag1")
+        assert karpv1.is_initialized(nc)
 
-    run(main())
+    run_spec(spec)
 
 
 def test_spec3_terminate_via_nodeclaim_delete():
-    async def main():
-        h = env()
-        await h.start()
-        try:
-            await h.kube.create(spec_nodeclaim("del1", {karpv1.KAITO_WORKSPACE_LABEL_KEY: "x"}))
-            nc = await h.wait_initialized("del1")
-            node_name = nc["status"]["nodeName"]
-            await h.kube.delete(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "del1")
-            await h.wait_gone(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "del1")
-            await h.wait_gone("v1", "Node", node_name)
-            assert "del1" not in h.agent_pools.pools
-        finally:
-            await h.stop()
+    async def spec(env):
+        await env.create(env.nodeclaim("del1", {karpv1.KAITO_WORKSPACE_LABEL_KEY: "x"}))
+        nc = await env.wait_initialized("del1")
+        node_name = nc["status"]["nodeName"]
+        await env.kube.delete(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "del1")
+        await env.wait_gone(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "del1")
+        await env.wait_gone("v1", "Node", node_name)
+        await env.expect_pool_gone("del1")
 
-    run(main())
+    run_spec(spec)
 
 
 def test_spec4_terminate_via_node_delete():
-    async def main():
-        h = env()
-        await h.start()
-        try:
-            await h.kube.create(spec_nodeclaim("del2", {karpv1.KAITO_WORKSPACE_LABEL_KEY: "x"}))
-            nc = await h.wait_initialized("del2")
-            node_name = nc["status"]["nodeName"]
-            await h.kube.delete("v1", "Node", node_name)
-            await h.wait_gone("v1", "Node", node_name)
-            await h.wait_gone(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "del2")
-            assert "del2" not in h.agent_pools.pools
-        finally:
-            await h.stop()
+    async def spec(env):
+        await env.create(env.nodeclaim("del2", {karpv1.KAITO_WORKSPACE_LABEL_KEY: "x"}))
+        nc = await env.wait_initialized("del2")
+        node_name = nc["status"]["nodeName"]
+        await env.kube.delete("v1", "Node", node_name)
+        await env.wait_gone("v1", "Node", node_name)
+        await env.wait_gone(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "del2")
+        await env.expect_pool_gone("del2")
 
-    run(main())
+    run_spec(spec)
 
 
 def test_spec5_provision_via_kaitonodeclass_ref():
     """Managed purely by NodeClassRef GroupKind, no kaito labels."""
 
-    async def main():
-        h = env()
-        await h.start()
+    async def spec(env):
         try:
-            await h.kube.create(await_class(h))
-            await h.kube.create(
-                spec_nodeclaim(
-                    "ncref1",
-                    labels={"app": "custom"},
-                    node_class=v1alpha1.node_class_ref("default"),
-                )
+            await env.create(v1alpha1.new_kaitonodeclass("default"))
+        except Exception:
+            pass  # live cluster may already carry the class
+        await env.create(
+            env.nodeclaim(
+                "ncref1",
+                labels={"app": "custom"},
+                node_class=v1alpha1.node_class_ref("default"),
             )
-            nc = await h.wait_initialized("ncref1")
-            assert karpv1.is_initialized(nc)
-        finally:
-            await h.stop()
+        )
+        nc = await env.wait_initialized("ncref1")
+        assert karpv1.is_initialized(nc)
 
-    run(main())
-
-
-def await_class(h):
-    return v1alpha1.new_kaitonodeclass("default")
+    run_spec(spec)
 
 
 def test_spec6_negative_foreign_nodeclass_ignored():
-    """AKSNodeClass ref + no kaito labels → no finalizer, no instance, no node
-    (reference suite_test.go:387-450)."""
+    """AKSNodeClass ref + no kaito labels → no finalizer, no instance, no
+    node (reference suite_test.go:387-450)."""
 
-    async def main():
-        h = env()
-        await h.start()
-        try:
-            await h.kube.create(
-                spec_nodeclaim(
-                    "foreign1",
-                    labels={"app": "other"},
-                    node_class={"group": "karpenter.azure.com", "kind": "AKSNodeClass", "name": "d"},
-                )
+    async def spec(env):
+        before = env.create_calls()
+        await env.create(
+            env.nodeclaim(
+                "foreign1",
+                labels={"app": "other"},
+                node_class={
+                    "group": "karpenter.azure.com",
+                    "kind": "AKSNodeClass",
+                    "name": "d",
+                },
             )
-            await asyncio.sleep(0.5)
-            nc = await h.kube.get(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "foreign1")
-            assert not ko.has_finalizer(nc, karpv1.TERMINATION_FINALIZER)
-            assert not karpv1.is_launched(nc)
-            assert h.agent_pools.create_calls == 0
-            assert len(await h.kube.list("v1", "Node")) == 0
-        finally:
-            await h.stop()
+        )
+        await asyncio.sleep(15.0 if env.is_live else 0.5)
+        nc = await env.kube.get(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "foreign1")
+        assert not ko.has_finalizer(nc, karpv1.TERMINATION_FINALIZER)
+        assert not karpv1.is_launched(nc)
+        after = env.create_calls()
+        if before is not None and after is not None:
+            assert after == before
+        # cleanup (no finalizer → immediate)
+        await env.kube.delete(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "foreign1")
 
-    run(main())
+    run_spec(spec)
 
 
 def test_spec7_azurelinux_annotation_sets_os_image():
-    async def main():
-        h = env()
-        await h.start()
-        try:
-            await h.kube.create(
-                spec_nodeclaim(
-                    "azlinux1",
-                    labels={karpv1.KAITO_WORKSPACE_LABEL_KEY: "x"},
-                    annotations={karpv1.NODE_IMAGE_FAMILY_ANNOTATION_KEY: "AzureLinux"},
-                )
+    async def spec(env):
+        await env.create(
+            env.nodeclaim(
+                "azlinux1",
+                labels={karpv1.KAITO_WORKSPACE_LABEL_KEY: "x"},
+                annotations={karpv1.NODE_IMAGE_FAMILY_ANNOTATION_KEY: "AzureLinux"},
             )
-            nc = await h.wait_initialized("azlinux1")
-            assert h.agent_pools.pools["azlinux1"]["properties"]["osSKU"] == "AzureLinux"
-            node = await h.kube.get("v1", "Node", nc["status"]["nodeName"])
-            assert "AzureLinux" in node["status"]["nodeInfo"]["osImage"]
-        finally:
-            await h.stop()
+        )
+        nc = await env.wait_initialized("azlinux1")
+        props = env.pool_properties("azlinux1")
+        if props is not None:
+            assert props["osSKU"] == "AzureLinux"
+        node = await env.kube.get("v1", "Node", nc["status"]["nodeName"])
+        os_image = node["status"]["nodeInfo"]["osImage"]
+        assert any(
+            s in os_image for s in ("AzureLinux", "Azure Linux", "CBL-Mariner", "Mariner")
+        ), os_image
 
-    run(main())
+    run_spec(spec)
 
 
 def test_spec8_delete_while_provisioning():
     """Deleting a NodeClaim mid-create must still tear everything down (the
     GC-covered crash window, reference delete-trigger spec + GC readme)."""
 
-    async def main():
-        h = Harness(create_latency=0.3, ready_latency=0.05).add_all_controllers(
-            gc_interval=0.3, adoption_age=0.1
-        )
-        await h.start()
-        try:
-            await h.kube.create(spec_nodeclaim("mid1", {karpv1.KAITO_WORKSPACE_LABEL_KEY: "x"}))
+    async def spec(env):
+        await env.create(env.nodeclaim("mid1", {karpv1.KAITO_WORKSPACE_LABEL_KEY: "x"}))
 
-            async def create_started():
-                return h.agent_pools.create_calls > 0 or None
+        async def create_started():
+            # backend-observable create counter, else the controller's
+            # finalizer ack — both mean "the create path is in flight"
+            calls = env.create_calls()
+            if calls is not None:
+                return True if calls > 0 else None
+            nc = await env.kube.get(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "mid1")
+            return True if ko.has_finalizer(nc, karpv1.TERMINATION_FINALIZER) else None
 
-            await h.wait_for(create_started)
-            # delete while the agent-pool LRO is still running
-            await h.kube.delete(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "mid1")
-            await h.wait_gone(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "mid1", timeout=20)
+        await env.eventually(create_started, desc="create started")
+        # delete while the agent-pool LRO is still running
+        await env.kube.delete(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "mid1")
+        await env.wait_gone(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "mid1")
+        await env.expect_pool_gone("mid1")
+        # no leaked nodes either
+        for node in await env.kube.list("v1", "Node"):
+            assert ko.labels_of(node).get(karpv1.AGENTPOOL_LABEL_KEY) != "mid1"
 
-            async def pool_gone():
-                return "mid1" not in h.agent_pools.pools or None
-
-            await h.wait_for(pool_gone, timeout=20)
-            # no leaked nodes either
-            for node in await h.kube.list("v1", "Node"):
-                assert ko.labels_of(node).get(karpv1.AGENTPOOL_LABEL_KEY) != "mid1"
-        finally:
-            await h.stop()
-
-    run(main())
+    run_spec(
+        spec,
+        create_latency=0.3,
+        ready_latency=0.05,
+        controllers={"gc_interval": 0.3, "adoption_age": 0.1},
+    )
 
 
+# ---------------------------------------------------- in-process-only tiers
+# (scripted simulator latency/faults — not reproducible on a live cluster)
+
+pytestmark_scale = pytest.mark.skipif(
+    LIVE, reason="scale/fault tiers script the simulator; live runs cover spec1-8"
+)
+
+
+def env_harness(**kw) -> Harness:
+    return Harness(ready_latency=0.05, plugin_latency=0.05, **kw).add_all_controllers(
+        gc_interval=60.0
+    )
+
+
+@pytestmark_scale
 def test_spec10_32_concurrent_churn_with_drift():
     """BASELINE config #5: 32 concurrent NodeClaims with drift detection +
     delete/disruption reconcile under churn. One third of the fleet is
@@ -270,6 +275,7 @@ def test_spec10_32_concurrent_churn_with_drift():
     run(main())
 
 
+@pytestmark_scale
 def test_spec11_concurrent_provisioning_pipelines():
     """8 concurrent NodeClaims against a cloud with real latency (0.3s LRO +
     0.05s node-ready) must provision in ~one latency budget, not 8 serial
@@ -304,12 +310,13 @@ def test_spec11_concurrent_provisioning_pipelines():
     run(main())
 
 
+@pytestmark_scale
 def test_spec9_workload_pod_binds_to_provisioned_node():
     """BASELINE config #4: a workload pod requesting amd.com/gpu schedules
     onto the provisioned MI355X node (binding simulated at the apiserver)."""
 
     async def main():
-        h = env()
+        h = env_harness()
         await h.start()
         try:
             await h.kube.create(spec_nodeclaim("bind1", {karpv1.KAITO_WORKSPACE_LABEL_KEY: "x"}))
@@ -339,6 +346,7 @@ def test_spec9_workload_pod_binds_to_provisioned_node():
     run(main())
 
 
+@pytestmark_scale
 def test_spec12_128_concurrent_churn_completes():
     """Regression for the teardown livelock: 128 concurrent NodeClaims
     through a full churn cycle must complete — sub-10ms requeue backstops
@@ -385,6 +393,7 @@ def test_spec12_128_concurrent_churn_completes():
     run(main(), timeout=200)
 
 
+@pytestmark_scale
 def test_spec13_512_claim_burst():
     """Burst scale: 512 concurrent NodeClaims (64 full MI355X hosts worth)
     provision and tear down without queue/worker degradation."""
@@ -430,6 +439,7 @@ def test_spec13_512_claim_burst():
     run(main(), timeout=240)
 
 
+@pytestmark_scale
 def test_metrics_wired_through_churn():
     """The karpenter_* metric series must actually move when the lifecycle
     acts — guards against silent metric rot."""
@@ -447,7 +457,7 @@ def test_metrics_wired_through_churn():
         created0 = counter_val(m.NODECLAIMS_CREATED, **labels)
         terminated0 = counter_val(m.NODECLAIMS_TERMINATED, **labels)
 
-        h = env()
+        h = env_harness()
         await h.start()
         try:
             await h.kube.create(spec_nodeclaim("met1", {karpv1.KAITO_WORKSPACE_LABEL_KEY: "w"}))
@@ -466,13 +476,14 @@ def test_metrics_wired_through_churn():
     run(main())
 
 
+@pytestmark_scale
 def test_spec14_spot_zone_pinned_claim():
     """The spot example's shape end-to-end: spot capacity type + zone
     pinning land on the pool (Spot priority, availabilityZones) and the
     cheapest eligible SKU is chosen."""
 
     async def main():
-        h = env()
+        h = env_harness()
         await h.start()
         try:
             nc = karpv1.new_nodeclaim(
