@@ -14,8 +14,9 @@ AZURE_CLUSTER_NAME ?= gpu-provisioner-amd-aks
 AZURE_LOCATION ?= eastus2
 IDENTITY_NAME ?= gpu-provisioner-amd-id
 
-.PHONY: all build nodeagent unit-test gpu-test e2etests bench lint clean \
-        docker-build helm-template az-mkrg az-mkaks az-identity az-federated-credential az-patch-helm
+.PHONY: all build nodeagent unit-test gpu-test e2etests bench lint lint-full clean \
+        docker-build docker-build-multiarch helm-template \
+        az-mkrg az-mkaks az-identity az-federated-credential az-patch-helm
 
 all: build
 
@@ -47,11 +48,26 @@ chaos-test:
 bench:
 	$(PYTHON) bench.py --steps 10 --warmup 3
 
+# offline tier: self-contained AST linter + bytecode compile; CI adds
+# ruff + mypy on top (.github/workflows/lint.yml). `make lint-full` runs
+# those too when they are installed locally.
 lint:
-	$(PYTHON) -m py_compile $$(git ls-files '*.py')
+	$(PYTHON) hack/lint.py
+	$(PYTHON) -m compileall -q gpu_provisioner_amd tests hack bench.py
+
+lint-full: lint
+	@command -v ruff >/dev/null && ruff check . || echo "ruff not installed; skipped (CI runs it)"
+	@command -v mypy >/dev/null && mypy gpu_provisioner_amd || echo "mypy not installed; skipped (CI runs it)"
 
 docker-build:
 	docker build --build-arg VERSION=$(VERSION) -t $(IMAGE) .
+
+# multi-arch (reference Makefile:127-160 docker-buildx): PUSH=--push to publish
+PLATFORMS ?= linux/amd64,linux/arm64
+PUSH ?=
+docker-build-multiarch:
+	docker buildx build --platform $(PLATFORMS) $(PUSH) \
+	  --build-arg VERSION=$(VERSION) -t $(IMAGE) .
 
 helm-template:
 	helm template gpu-provisioner-amd charts/gpu-provisioner-amd \

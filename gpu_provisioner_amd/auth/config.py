@@ -8,7 +8,7 @@ boot, so the reference's Helm values drive this controller unchanged.
 from __future__ import annotations
 
 import os
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 
 DEPLOYMENT_MODE_MANAGED = "managed"
 DEPLOYMENT_MODE_SELF_HOSTED = "self-hosted"

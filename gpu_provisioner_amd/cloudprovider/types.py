@@ -12,7 +12,7 @@ import abc
 from dataclasses import dataclass, field
 from typing import Optional
 
-from ..kube.objects import Quantity, qty
+from ..kube.objects import qty
 
 
 # ---------------------------------------------------------------------------

@@ -20,7 +20,6 @@ from __future__ import annotations
 
 import asyncio
 import logging
-from datetime import timedelta
 from typing import Optional
 
 from ...apis import v1 as karpv1

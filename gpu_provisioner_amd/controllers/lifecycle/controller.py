@@ -26,7 +26,6 @@ Design differences from the reference (deliberate):
 """
 from __future__ import annotations
 
-import asyncio
 import logging
 import time
 from collections import OrderedDict

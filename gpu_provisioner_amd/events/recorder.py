@@ -11,7 +11,6 @@ import logging
 import time
 import uuid
 from collections import OrderedDict
-from typing import Optional
 
 from ..kube import objects as ko
 from ..kube.client import KubeClient

@@ -30,9 +30,7 @@ from ..providers.instance.armapi import (
     AgentPoolsAPI,
     ARMError,
     LROPoller,
-    pool_labels,
     pool_name,
-    pool_state,
     taint_from_string,
 )
 from ..utils.utils import build_provider_id

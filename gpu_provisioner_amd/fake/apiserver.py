@@ -38,7 +38,6 @@ from ..kube.client import (
     KubeClient,
     LabelSelector,
     NotFoundError,
-    TooManyRequestsError,
     json_merge_patch,
     match_field_selector,
 )

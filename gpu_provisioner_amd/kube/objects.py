@@ -9,12 +9,11 @@ behavioral spec.
 """
 from __future__ import annotations
 
-import copy
 import re
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from fractions import Fraction
 from datetime import datetime, timezone
-from typing import Any, Iterable, Optional
+from typing import Iterable, Optional
 
 # ---------------------------------------------------------------------------
 # time helpers (RFC3339, as the apiserver emits)

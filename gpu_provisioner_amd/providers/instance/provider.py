@@ -40,7 +40,6 @@ from .armapi import (
     is_create_in_progress,
     pool_labels,
     pool_name,
-    pool_props,
     pool_state,
     pool_vm_size,
     taint_to_string,

@@ -4,12 +4,10 @@ VERDICT r01 #3: every agent-pool PUT body must validate against the pinned
 api-version's schema; speculative gpuProfile fields live behind the
 explicitly-selected gpu-preview profile; recorded fixtures exercise real ARM
 wire shapes (error bodies, header casing, api-version query)."""
-import json
 
 import httpx
 import pytest
 
-from gpu_provisioner_amd.apis import v1 as karpv1
 from gpu_provisioner_amd.auth.cred import StaticCredential
 from gpu_provisioner_amd.providers.instance import bootstrap
 from gpu_provisioner_amd.providers.instance.armclient import ARMAgentPoolsClient
@@ -20,8 +18,6 @@ from gpu_provisioner_amd.providers.instance.armschema import (
     profile_from_env,
     validate_agent_pool,
 )
-from gpu_provisioner_amd.providers.instance.provider import InstanceProvider
-from gpu_provisioner_amd.providers.instancetype.catalog import InstanceTypeProvider
 from tests.conftest import run
 from tests.test_instance_provider import make_provider, nodeclaim, VM
 

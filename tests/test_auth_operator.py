@@ -1,7 +1,6 @@
 """Auth config matrix (reference pkg/auth/config_test.go), credential flows
 (mocked AAD/IMDS), options/feature gates, logging format, leader election."""
 import asyncio
-import io
 import json
 import logging
 import time
@@ -10,7 +9,6 @@ import httpx
 import pytest
 
 from gpu_provisioner_amd.auth.config import (
-    AzureConfig,
     ConfigError,
     build_azure_config,
 )
@@ -21,7 +19,7 @@ from gpu_provisioner_amd.auth.cred import (
 )
 from gpu_provisioner_amd.fake.apiserver import InMemoryAPIServer, InMemoryClient
 from gpu_provisioner_amd.operator.leaderelection import LeaderElector
-from gpu_provisioner_amd.operator.logging import JSONFormatter, setup_logging
+from gpu_provisioner_amd.operator.logging import JSONFormatter
 from gpu_provisioner_amd.operator.options import FeatureGates, Options
 from tests.conftest import run
 

@@ -3,7 +3,6 @@ examples) must validate against the NodeClaim CRD's openAPIV3Schema. A
 minimal structural validator (type/properties/required/items/enum) is
 enough to catch drift between the Python surface and the CRD the chart
 installs."""
-import asyncio
 import os
 
 import pytest

@@ -5,7 +5,6 @@ nodeclaim.drift singleton maintains the Drifted condition, and the
 DriftReplace gate turns detection into replacement."""
 import asyncio
 
-import pytest
 
 from gpu_provisioner_amd.apis import v1 as karpv1
 from gpu_provisioner_amd.cloudprovider.azure import (
@@ -254,7 +253,6 @@ def test_feature_gates_parse_drift():
 
 def test_build_manager_registers_drift_controller():
     from gpu_provisioner_amd.controllers.drift.controller import DriftController
-    from gpu_provisioner_amd.fake.apiserver import InMemoryAPIServer, InMemoryClient
     from gpu_provisioner_amd.main import build_manager
     from gpu_provisioner_amd.operator.options import Options
 

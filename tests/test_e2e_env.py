@@ -3,7 +3,6 @@ kubeconfig-driven live client construction (E2E_LIVE
 must be real and must fail LOUDLY when misconfigured, never silently re-run
 the simulator)."""
 import base64
-import os
 
 import pytest
 import yaml

@@ -3,7 +3,6 @@ Behavioral spec: reference pkg/controllers/instance/garbagecollection/ and
 vendor/.../nodeclaim/garbagecollection/ (§3.4), node/health (§3.5)."""
 import asyncio
 
-import pytest
 
 from gpu_provisioner_amd.apis import v1 as karpv1
 from gpu_provisioner_amd.fake.harness import Harness

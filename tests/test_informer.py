@@ -5,7 +5,7 @@ import asyncio
 import pytest
 
 from gpu_provisioner_amd.fake.apiserver import InMemoryAPIServer, InMemoryClient
-from gpu_provisioner_amd.kube.client import ADDED, DELETED, GoneError, MODIFIED
+from gpu_provisioner_amd.kube.client import ADDED, DELETED, MODIFIED
 from gpu_provisioner_amd.kube.informer import Informer, InformerFactory
 from tests.conftest import run
 

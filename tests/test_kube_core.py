@@ -10,7 +10,6 @@ from gpu_provisioner_amd.kube import objects as ko
 from gpu_provisioner_amd.kube.client import (
     ADDED,
     DELETED,
-    MODIFIED,
     AlreadyExistsError,
     ConflictError,
     LabelSelector,

@@ -5,11 +5,9 @@ scenarios of the reference's lifecycle controller
 e2e suite provision/terminate specs (test/e2e/suites/suite_test.go)."""
 import asyncio
 
-import pytest
 
 from gpu_provisioner_amd.apis import v1 as karpv1
 from gpu_provisioner_amd.controllers.lifecycle.controller import LifecycleController
-from gpu_provisioner_amd.cloudprovider.types import InsufficientCapacityError
 from gpu_provisioner_amd.fake.harness import Harness
 from gpu_provisioner_amd.kube import objects as ko
 from gpu_provisioner_amd.providers.instance.armapi import ARMError

@@ -5,7 +5,6 @@ import asyncio
 import pytest
 
 from gpu_provisioner_amd.apis import v1 as karpv1
-from gpu_provisioner_amd.cloudprovider.types import InstanceType, Offering
 from gpu_provisioner_amd.events.recorder import EventRecorder
 from gpu_provisioner_amd.fake.apiserver import InMemoryAPIServer, InMemoryClient
 from gpu_provisioner_amd.kube import objects as ko

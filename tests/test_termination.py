@@ -3,7 +3,6 @@ volume-detach wait, finalizer interplay with the NodeClaim lifecycle.
 Behavioral spec: reference vendor/.../controllers/node/termination/."""
 import asyncio
 
-import pytest
 
 from gpu_provisioner_amd.apis import v1 as karpv1
 from gpu_provisioner_amd.fake.harness import Harness
