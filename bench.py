@@ -79,19 +79,35 @@ async def one_step(h: Harness, step: int, concurrent: int, latencies: list) -> N
 async def run_bench(
     steps: int, warmup: int, concurrent: int,
     create_latency: float = 0.0, ready_latency: float = 0.0,
+    rss_every: int = 0,
 ) -> dict:
     h = build_harness(create_latency, ready_latency)
     await h.start()
     try:
+        rss: list = []
+
+        def sample_rss(step: int) -> None:
+            # soak-memory evidence: flat RSS over long
+            # runs is the leak check the reference covers with its heap
+            # profile endpoint
+            if rss_every and step % rss_every == 0:
+                import psutil
+
+                rss.append(
+                    {"step": step, "rss_bytes": psutil.Process().memory_info().rss}
+                )
+
         warm_lat: list = []
         for s in range(warmup):
             await one_step(h, s, concurrent, warm_lat)
         latencies: list = []
         t0 = time.monotonic()
         for s in range(warmup, warmup + steps):
+            sample_rss(s)
             await one_step(h, s, concurrent, latencies)
+        sample_rss(warmup + steps)
         elapsed = time.monotonic() - t0
-        return {"elapsed_s": elapsed, "latencies": latencies}
+        return {"elapsed_s": elapsed, "latencies": latencies, "rss": rss}
     finally:
         await h.stop()
 
@@ -107,6 +123,10 @@ def main() -> None:
                     help="simulated agent-pool LRO seconds (0 = BASELINE config #1)")
     ap.add_argument("--ready-latency", type=float, default=0.0,
                     help="simulated node boot-to-Ready seconds")
+    ap.add_argument("--rss-out", default="",
+                    help="write per-step RSS samples (soak memory evidence) to this JSON file")
+    ap.add_argument("--rss-every", type=int, default=100,
+                    help="sample RSS every N steps when --rss-out is set")
     args = ap.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -136,9 +156,22 @@ def main() -> None:
         run_bench(
             args.steps, args.warmup, args.concurrent,
             args.create_latency, args.ready_latency,
+            rss_every=(args.rss_every if args.rss_out else 0),
         )
     )
     sync()
+    if args.rss_out and rank == 0:
+        rss = result.get("rss", [])
+        summary = {
+            "steps": args.steps,
+            "concurrent": args.concurrent,
+            "samples": rss,
+            "rss_first_bytes": rss[0]["rss_bytes"] if rss else None,
+            "rss_last_bytes": rss[-1]["rss_bytes"] if rss else None,
+            "rss_max_bytes": max((s["rss_bytes"] for s in rss), default=None),
+        }
+        with open(args.rss_out, "w") as f:
+            json.dump(summary, f, indent=1)
 
     elapsed = result["elapsed_s"]
     if dist is not None:

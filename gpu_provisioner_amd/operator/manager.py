@@ -70,7 +70,12 @@ class Manager:
 
         routes = [Route("/metrics", metrics)]
         if self.options.enable_profiling:
-            # pprof-equivalent debug endpoints (reference operator.go:181-197)
+            # pprof-equivalent debug endpoints. The reference exposes
+            # heap/block/goroutine/profile/trace via net/http/pprof
+            # (vendor/.../operator/operator.go:181-197); the Python
+            # analogues: tracemalloc (heap/allocs), per-thread stacks
+            # (goroutine), cProfile over a window (profile), task dump
+            # (trace-ish) and an event-loop lag sampler (block).
             async def stacks(request):
                 frames = sys._current_frames()
                 out = {}
@@ -83,8 +88,87 @@ class Manager:
                     [repr(t) for t in asyncio.all_tasks()], status_code=200
                 )
 
+            async def heap(request):
+                import tracemalloc
+
+                if not tracemalloc.is_tracing():
+                    return PlainTextResponse(
+                        "tracemalloc not tracing (profiling just enabled?)",
+                        status_code=503,
+                    )
+                limit = int(request.query_params.get("n", "50"))
+                snap = tracemalloc.take_snapshot()
+                stats = snap.statistics(
+                    "traceback" if request.query_params.get("traceback") else "lineno"
+                )
+                current, peak = tracemalloc.get_traced_memory()
+                return JSONResponse(
+                    {
+                        "traced_current_bytes": current,
+                        "traced_peak_bytes": peak,
+                        "top": [
+                            {
+                                "site": str(s.traceback),
+                                "size_bytes": s.size,
+                                "count": s.count,
+                            }
+                            for s in stats[:limit]
+                        ],
+                    }
+                )
+
+            async def profile(request):
+                """cProfile the process for ?seconds=N (default 5) and
+                return the top cumulative entries as text."""
+                import cProfile
+                import io
+                import pstats
+
+                seconds = min(float(request.query_params.get("seconds", "5")), 60.0)
+                prof = cProfile.Profile()
+                prof.enable()
+                await asyncio.sleep(seconds)
+                prof.disable()
+                buf = io.StringIO()
+                pstats.Stats(prof, stream=buf).sort_stats("cumulative").print_stats(60)
+                return PlainTextResponse(buf.getvalue())
+
+            async def block(request):
+                """Event-loop lag sampled over ?seconds=N — the asyncio
+                analogue of the block profile: how long ready callbacks
+                wait for the loop."""
+                seconds = min(float(request.query_params.get("seconds", "5")), 60.0)
+                import time as _time
+
+                samples: list = []
+                deadline = _time.monotonic() + seconds
+                while _time.monotonic() < deadline:
+                    t0 = _time.monotonic()
+                    await asyncio.sleep(0.01)
+                    samples.append(max(0.0, _time.monotonic() - t0 - 0.01))
+                samples.sort()
+
+                def pct(p):
+                    return round(samples[min(len(samples) - 1, int(len(samples) * p))] * 1000, 3)
+
+                return JSONResponse(
+                    {
+                        "samples": len(samples),
+                        "loop_lag_ms": {
+                            "p50": pct(0.50),
+                            "p90": pct(0.90),
+                            "p99": pct(0.99),
+                            "max": round(samples[-1] * 1000, 3) if samples else 0,
+                        },
+                    }
+                )
+
             routes += [
                 Route("/debug/pprof/goroutine", stacks),
+                Route("/debug/pprof/heap", heap),
+                Route("/debug/pprof/allocs", heap),
+                Route("/debug/pprof/profile", profile),
+                Route("/debug/pprof/block", block),
                 Route("/debug/tasks", tasks),
             ]
         return Starlette(routes=routes)
@@ -119,6 +203,14 @@ class Manager:
 
     async def start(self, serve_http: bool = True) -> None:
         BUILD_INFO.labels(version=self.version).set(1)
+        if self.options.enable_profiling:
+            # heap endpoint needs allocation tracing from process start-ish;
+            # 10 frames keeps per-alloc overhead modest while still
+            # attributing sites usefully
+            import tracemalloc
+
+            if not tracemalloc.is_tracing():
+                tracemalloc.start(10)
         if serve_http:
             self._server_tasks = [
                 asyncio.create_task(

@@ -111,6 +111,26 @@ def test_graceful_stop_releases_lease_for_fast_takeover():
             takeover = time.monotonic() - t0
             # must be driven by the release, not the 30s lease expiry
             assert takeover < 10, f"takeover took {takeover:.1f}s"
+            # HA metric artifact: the measured graceful-failover takeover
+            # time, recorded alongside the scaling bench when HA_METRIC_OUT
+            # is set
+            import json
+            import os
+
+            out = os.environ.get("HA_METRIC_OUT")
+            if out:
+                with open(out, "w") as f:
+                    json.dump(
+                        {
+                            "metric": "leader_failover_takeover_s",
+                            "value": round(takeover, 4),
+                            "mode": "graceful (lease released on stop)",
+                            "lease_duration_s": 30.0,
+                            "higher_is_better": False,
+                        },
+                        f,
+                        indent=1,
+                    )
         finally:
             await m2.stop()
 

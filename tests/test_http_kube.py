@@ -221,6 +221,31 @@ def test_profiling_debug_endpoints():
         assert resp.status_code == 200 and resp.json()  # per-thread stacks
         resp = await app.get("/debug/tasks")
         assert resp.status_code == 200
+        # heap/allocs (tracemalloc) — 503 until tracing starts, live after
+        import tracemalloc
+
+        was_tracing = tracemalloc.is_tracing()
+        if not was_tracing:
+            resp = await app.get("/debug/pprof/heap")
+            assert resp.status_code == 503
+            tracemalloc.start(5)
+        try:
+            resp = await app.get("/debug/pprof/heap?n=5")
+            body = resp.json()
+            assert resp.status_code == 200
+            assert body["traced_current_bytes"] > 0
+            assert len(body["top"]) <= 5 and body["top"][0]["size_bytes"] > 0
+            resp = await app.get("/debug/pprof/allocs")
+            assert resp.status_code == 200
+        finally:
+            if not was_tracing:
+                tracemalloc.stop()
+        # cpu profile + loop-lag (block) over a short window
+        resp = await app.get("/debug/pprof/profile?seconds=0.2")
+        assert resp.status_code == 200 and b"cumulative" in resp.content
+        resp = await app.get("/debug/pprof/block?seconds=0.2")
+        lag = resp.json()["loop_lag_ms"]
+        assert resp.status_code == 200 and lag["max"] >= 0
         # without the flag the debug routes are absent
         mgr2 = Manager(kube, Options.from_env_and_args([], {}))
         app2 = httpx.AsyncClient(
