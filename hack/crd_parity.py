@@ -145,7 +145,7 @@ def compare(ref_path: str, repo_path: str) -> dict:
 
 def render_markdown(result: dict, ref_path: str) -> str:
     lines = [
-        "# PARITY — NodeClaim CRD vs reference",
+        "# CRD PARITY — NodeClaim CRD vs reference",
         "",
         f"Structural validation-feature diff of `{os.path.relpath(REPO_CRD, REPO_ROOT)}`",
         f"against `{ref_path}`, produced by `hack/crd_parity.py` (run in",
@@ -206,7 +206,7 @@ def main() -> int:
     md = render_markdown(result, ref_path)
     print(md)
     if args.write_parity_md:
-        with open(os.path.join(REPO_ROOT, "PARITY.md"), "w") as f:
+        with open(os.path.join(REPO_ROOT, "CRD_PARITY.md"), "w") as f:
             f.write(md)
     ok = (
         not result["missing"]
