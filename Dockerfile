@@ -3,7 +3,14 @@
 # The controller itself needs no ROCm — only nodes run the HIP agent — so
 # the runtime stage stays small and the nodeagent DaemonSet reuses this
 # image on GPU nodes where /dev/kfd exists.
-FROM rocm/dev-ubuntu-22.04:6.4 AS builder
+#
+# Multi-arch: the HIP builder stage is pinned to amd64 (ROCm toolchain)
+# — under buildx it builds the gfx950 code object once and every target
+# arch copies it in. MI355X nodes are amd64, so the arm64 image variant
+# (control-plane-only deployments) carries the library as inert payload;
+# the node agent only ever executes on amd64 GPU nodes and fails loudly
+# anywhere the library cannot load.
+FROM --platform=linux/amd64 rocm/dev-ubuntu-22.04:6.4 AS builder
 WORKDIR /src
 COPY nodeagent/ nodeagent/
 RUN hipcc --offload-arch=gfx950 -O3 -shared -fPIC \

@@ -345,9 +345,14 @@ class LifecycleController:
                 ko.node_taints(node), nodeclaim.get("spec", {}).get("taints") or []
             )
             node_finalizers = list(ko.finalizers_of(node))
-            if karpv1.TERMINATION_FINALIZER not in node_finalizers:
+            if (
+                karpv1.TERMINATION_FINALIZER not in node_finalizers
+                and not ko.is_deleting(node)
+            ):
                 # the node carries the termination finalizer so node deletion
-                # runs the drain pipeline before the kubelet object vanishes
+                # runs the drain pipeline before the kubelet object vanishes.
+                # Never append to a TERMINATING node: a real apiserver
+                # rejects new finalizers on deleting objects with 422.
                 node_finalizers.append(karpv1.TERMINATION_FINALIZER)
             patch: dict = {
                 "metadata": {

@@ -246,6 +246,17 @@ class InMemoryAPIServer:
             stored = {**cur, "metadata": {**cur["metadata"]}}
             stored["status"] = ko.deep_copy(obj.get("status", {}))
         else:
+            # real-apiserver rule (registry/rest validation): once an object
+            # is terminating, NEW finalizers may not be added — attempting
+            # it is a 422 "no new finalizers can be added if the object is
+            # being deleted"
+            if ko.is_deleting(cur):
+                added = set(ko.finalizers_of(obj)) - set(ko.finalizers_of(cur))
+                if added:
+                    raise InvalidError(
+                        "metadata.finalizers: Forbidden: no new finalizers can "
+                        f"be added if the object is being deleted (added: {sorted(added)})"
+                    )
             new = ko.deep_copy(obj)
             # status is a subresource: ignore status changes on main-resource update
             new["status"] = cur.get("status", {})

@@ -232,3 +232,30 @@ def test_controllers_never_mutate_informer_cache_in_place():
             await h.stop()
 
     run(main())
+
+
+def test_no_new_finalizers_on_terminating_object():
+    """Real apiservers reject adding finalizers to a deleting object with
+    422 (registry/rest update validation: 'no new finalizers can be added
+    if the object is being deleted'); removing finalizers stays allowed."""
+
+    async def main():
+        kube = client()
+        nc = karpv1.new_nodeclaim("fin1", labels={})
+        nc["metadata"]["finalizers"] = ["keep.example.com/one"]
+        await kube.create(nc)
+        await kube.delete(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "fin1")
+        cur = await kube.get(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "fin1")
+        assert ko.is_deleting(cur)
+        # adding → 422
+        cur["metadata"]["finalizers"] = ["keep.example.com/one", "new.example.com/two"]
+        with pytest.raises(InvalidError, match="no new finalizers"):
+            await kube.update(cur)
+        # removing → allowed, and removing the last one completes deletion
+        cur = await kube.get(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "fin1")
+        cur["metadata"]["finalizers"] = []
+        await kube.update(cur)
+        with pytest.raises(Exception):
+            await kube.get(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "fin1")
+
+    run(main())
