@@ -94,6 +94,9 @@ class LifecycleController:
         # not be trusted — re-running the chain on pre-write state costs a
         # stale pass per write (measured ~20% bench throughput).
         self._written_rv: OrderedDict = OrderedDict()
+        # echo-suppression floor for our own NODE writes (registration
+        # taint/label sync, initialization label): node name -> rv
+        self._node_written_rv: OrderedDict = OrderedDict()
         self.controller = Controller(
             self.NAME,
             self.reconcile,
@@ -122,14 +125,54 @@ class LifecycleController:
             self._written_rv.popitem(last=False)
 
     def _on_nodeclaim_event(self, event_type: str, obj: dict) -> None:
+        name = ko.name_of(obj)
         if event_type == "DELETED":
-            self._written_rv.pop(ko.name_of(obj), None)
+            self._written_rv.pop(name, None)
+        elif event_type == "MODIFIED":
+            # echo suppression: a MODIFIED event at-or-below the rv of OUR
+            # OWN last write is the watch echoing that write back — the
+            # reconcile that performed it already ran the full chain after
+            # the write, so the pass would be a no-op. Foreign writes carry
+            # a higher rv and still enqueue; progress otherwise rides on
+            # returned requeues and foreign events. Measured: ~1/3 of all
+            # lifecycle reconcile passes were self-echoes.
+            floor = self._written_rv.get(name)
+            if floor is not None:
+                try:
+                    if int(ko.meta(obj).get("resourceVersion") or 0) <= floor:
+                        return
+                except (TypeError, ValueError):
+                    pass
         if karpv1.is_managed(obj):
-            self.controller.enqueue_nowait(ko.name_of(obj))
+            self.controller.enqueue_nowait(name)
+
+    def _record_node_write(self, node_name: str, obj: dict) -> None:
+        try:
+            rv = int(ko.meta(obj).get("resourceVersion") or 0)
+        except (TypeError, ValueError):
+            return
+        self._node_written_rv[node_name] = max(self._node_written_rv.get(node_name, 0), rv)
+        self._node_written_rv.move_to_end(node_name)
+        while len(self._node_written_rv) > 4096:
+            self._node_written_rv.popitem(last=False)
 
     def _on_node_event(self, event_type: str, obj: dict) -> None:
         """Map Node events to their NodeClaim via providerID index
         (reference controller.go:92-108)."""
+        node_name = ko.name_of(obj)
+        if event_type == "DELETED":
+            self._node_written_rv.pop(node_name, None)
+        elif event_type == "MODIFIED":
+            # echo suppression, same argument as _on_nodeclaim_event: the
+            # reconcile that patched this node kept running its chain after
+            # the write, so the watch echo adds nothing
+            floor = self._node_written_rv.get(node_name)
+            if floor is not None:
+                try:
+                    if int(ko.meta(obj).get("resourceVersion") or 0) <= floor:
+                        return
+                except (TypeError, ValueError):
+                    pass
         pid = ko.provider_id_of(obj)
         if not pid:
             return
@@ -288,6 +331,13 @@ class LifecycleController:
                 status[f] = created_status[f]
         ko.set_condition(nodeclaim, karpv1.COND_LAUNCHED, ko.CONDITION_TRUE, "Launched")
         await self._patch_status(nodeclaim)
+        # the cache's only job is re-launch dedup while Launched is not yet
+        # persisted; once the status write lands (read-your-writes floor
+        # guarantees later reconciles see it) the entry — holding the full
+        # created-instance tree — is dead weight. Dropping it here turns a
+        # rate×TTL memory window (~40k entries at bench churn) into
+        # in-flight-only; the TTL sweep stays as the crash-path backstop.
+        self._launch_cache.pop(uid, None)
         if launched_now:
             # a stale cached-client read can re-run this sub-reconciler after
             # the claim is already Launched; the UID cache dedupes the cloud
@@ -372,7 +422,8 @@ class LifecycleController:
                 "spec": {"taints": node_taints or None},
             }
             try:
-                await self.kube.patch("v1", "Node", ko.name_of(node), patch)
+                updated_node = await self.kube.patch("v1", "Node", ko.name_of(node), patch)
+                self._record_node_write(ko.name_of(node), updated_node)
                 break
             except ConflictError:
                 try:
@@ -406,12 +457,13 @@ class LifecycleController:
             ):
                 await self._patch_status(nodeclaim)
             return Result(requeue_after=REGISTRATION_REQUEUE)
-        await self.kube.patch(
+        updated_node = await self.kube.patch(
             "v1",
             "Node",
             ko.name_of(node),
             {"metadata": {"labels": {karpv1.NODE_INITIALIZED_LABEL_KEY: "true"}}},
         )
+        self._record_node_write(ko.name_of(node), updated_node)
         status = nodeclaim.setdefault("status", {})
         # `node` can be the SHARED informer-cache object (_nodes_by_provider_id
         # returns it uncopied); aliasing its capacity/allocatable subtrees into

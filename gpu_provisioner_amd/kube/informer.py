@@ -311,7 +311,15 @@ class Informer:
                 vals = [vals]
             for v in vals:
                 if remove:
-                    idx[v].discard(key)
+                    s = idx.get(v)
+                    if s is not None:
+                        s.discard(key)
+                        # drop emptied buckets: a defaultdict(set) keeps one
+                        # empty set per index value EVER seen, which is an
+                        # unbounded live-object leak at churn (measured
+                        # ~5 sets/claim, +260 MB RSS over a 20k-step soak)
+                        if not s:
+                            del idx[v]
                 else:
                     idx[v].add(key)
 
