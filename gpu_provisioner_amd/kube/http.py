@@ -223,7 +223,10 @@ class HTTPClient(KubeClient):
             raise GoneError(message)
         if code == 403:
             raise ForbiddenError(message)
-        if code == 422:
+        if code in (400, 422):
+            # 400 BadRequest (e.g. "field label not supported" selectors)
+            # and 422 Invalid (schema validation) both mean the request
+            # itself is wrong — retrying is pointless, surface as Invalid
             raise InvalidError(message)
         if code == 429:
             retry = float(resp.headers.get("Retry-After", "1"))
