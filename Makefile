@@ -34,9 +34,15 @@ unit-test:
 gpu-test: nodeagent
 	$(PYTHON) -m pytest tests/ -q -m gpu
 
-# the in-process e2e suite (full controller topology against the AKS simulator)
+# the e2e suite: in-process backend, then the same 8 reference specs over
+# the production HTTP transport (envtest-lite REST server)
 e2etests:
 	$(PYTHON) -m pytest tests/test_e2e_suite.py -q
+	E2E_TRANSPORT=http $(PYTHON) -m pytest tests/test_e2e_suite.py -q
+
+# live-cluster backend: requires E2E_LIVE=1 + KUBECONFIG (fails loudly without)
+e2etests-live:
+	E2E_LIVE=1 $(PYTHON) -m pytest tests/test_e2e_suite.py -q --timeout 3600
 
 # reliability tiers: chaos (CHAOS_SEED=N sweeps interleavings), crash-restart,
 # HA failover, workqueue/informer model checks

@@ -353,9 +353,83 @@ class LiveEnv(E2EEnvironment):
                 print(f"(log fetch failed: {e})")
 
 
+class HttpInProcessEnv(E2EEnvironment):
+    """In-process controllers + AKS simulator, but the controllers AND the
+    specs speak to the apiserver through the PRODUCTION HTTP transport
+    against fake/restserver.py over 127.0.0.1 (the envtest-lite wire
+    layer). Select with E2E_TRANSPORT=http."""
+
+    def __init__(self, **harness_kw):
+        from gpu_provisioner_amd.cloudprovider.azure import AzureCloudProvider
+        from gpu_provisioner_amd.fake.agentpools import AKSSimulator, FakeAgentPools
+        from gpu_provisioner_amd.fake.apiserver import InMemoryAPIServer, InMemoryClient
+        from gpu_provisioner_amd.fake.restserver import RESTServerHandle
+        from gpu_provisioner_amd.providers.instance.provider import InstanceProvider
+        from gpu_provisioner_amd.providers.instancetype.catalog import InstanceTypeProvider
+
+        self._harness_kw = harness_kw  # latency knobs
+        self.server = InMemoryAPIServer()
+        self.rest = RESTServerHandle(self.server)
+        self.actor_client = InMemoryClient(self.server)
+        self.catalog = InstanceTypeProvider()
+        self.pools = FakeAgentPools(
+            create_latency=harness_kw.get("create_latency", 0.0),
+            delete_latency=harness_kw.get("delete_latency", 0.0),
+        )
+        self.aks = AKSSimulator(
+            self.actor_client,
+            self.pools,
+            ready_latency=harness_kw.get("ready_latency", 0.05),
+            plugin_latency=harness_kw.get("plugin_latency", 0.05),
+            gpu_count_for=self.catalog.gpu_count,
+        )
+        self._azure_cloud = None
+        self.manager = None
+        self.run_id = f"http-{uuid.uuid4().hex[:8]}"
+        self.default_timeout = 30.0
+
+    async def start(self) -> None:
+        from gpu_provisioner_amd.cloudprovider.azure import AzureCloudProvider
+        from gpu_provisioner_amd.kube.http import HTTPClient
+        from gpu_provisioner_amd.main import build_manager
+        from gpu_provisioner_amd.operator.options import Options
+        from gpu_provisioner_amd.providers.instance.provider import InstanceProvider
+
+        port = await self.rest.start()
+        self.kube = HTTPClient(f"http://127.0.0.1:{port}")
+        instances = InstanceProvider(
+            self.pools, self.kube, self.catalog, "rg", "cluster",
+            node_wait_interval=0.02,
+        )
+        cloud = AzureCloudProvider(instances, self.catalog)
+        self.manager = build_manager(self.kube, Options(), cloud)
+        await self.manager.start(serve_http=False)
+
+    async def stop(self) -> None:
+        if self.manager is not None:
+            await self.manager.stop()
+        await self.kube.close()
+        await self.rest.stop()
+
+    def pool_properties(self, name: str) -> Optional[dict]:
+        pool = self.pools.pools.get(name)
+        return pool.get("properties") if pool else None
+
+    def pool_exists(self, name: str) -> Optional[bool]:
+        return name in self.pools.pools
+
+    def create_calls(self) -> Optional[int]:
+        return self.pools.create_calls
+
+
 def make_env(**in_process_kw) -> E2EEnvironment:
     """Backend selector. E2E_LIVE=1 → LiveEnv (misconfiguration FAILS);
-    default → InProcessEnv."""
+    E2E_TRANSPORT=http → HttpInProcessEnv (production HTTP transport over
+    the envtest-lite REST server); default → InProcessEnv."""
     if os.environ.get("E2E_LIVE", "") == "1":
         return LiveEnv()
+    if os.environ.get("E2E_TRANSPORT", "") == "http":
+        kw = dict(in_process_kw)
+        kw.pop("controllers", None)  # manager wiring uses production cadences
+        return HttpInProcessEnv(**kw)
     return InProcessEnv(**in_process_kw)
