@@ -123,6 +123,11 @@ class InMemoryAPIServer:
         self._history: deque = deque(maxlen=_WATCH_HISTORY)  # (rv, gvk, event_type, obj)
         # test hooks: fn(verb, gvk, obj_or_name) -> Optional[APIError] raised if returned
         self.reactors: list = []
+        # server-side CRD schema validation (real-apiserver behavior):
+        # gvk -> fn(new, old_or_None) -> list[str]; violations → 422
+        # (kube/crdschema.CRDValidator; the harness registers the chart's
+        # NodeClaim/KaitoNodeClass CRDs here)
+        self.validators: dict = {}
         # eviction hook: fn(pod) -> Optional[APIError]
         self.eviction_reactor: Optional[Callable] = None
         self.evictions: list = []  # recorded (namespace, name)
@@ -152,6 +157,16 @@ class InMemoryAPIServer:
 
     def _bump(self, obj: dict) -> None:
         obj["metadata"]["resourceVersion"] = self._next_rv()
+
+    def _validate(self, gvk: tuple, new: dict, old: Optional[dict]) -> None:
+        fn = self.validators.get(gvk)
+        if fn is None:
+            return
+        errs = fn(new, old)
+        if errs:
+            raise InvalidError(
+                f"{gvk[1]} {ko.name_of(new)} is invalid: " + "; ".join(errs[:5])
+            )
 
     # -- verbs (all called under the lock by InMemoryClient) ---------------
 
@@ -210,6 +225,7 @@ class InMemoryAPIServer:
             m["uid"] = m.get("uid") or str(uuid.uuid4())
             m["creationTimestamp"] = m.get("creationTimestamp") or ko.fmt_time(ko.now())
             m["generation"] = 1
+            self._validate(gvk, stored, None)  # CRD schema + defaults (422)
             self._bump(stored)
             self._store[gvk][key] = stored
             # Events are capped like a real cluster's event TTL would bound
@@ -271,6 +287,7 @@ class InMemoryAPIServer:
             if new.get("spec") != cur.get("spec"):
                 nm["generation"] = cm.get("generation", 1) + 1
             stored = new
+        self._validate(gvk, stored, cur)  # CRD schema + CEL immutability (422)
         self._bump(stored)
         # finalizer-aware deletion: removing last finalizer on a deleting object
         if ko.is_deleting(stored) and not ko.finalizers_of(stored) and subresource != "status":

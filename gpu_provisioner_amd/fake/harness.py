@@ -24,6 +24,30 @@ from .agentpools import AKSSimulator, FakeAgentPools
 from .apiserver import InMemoryAPIServer, InMemoryClient
 
 
+def install_chart_crd_validators(server: InMemoryAPIServer) -> None:
+    """Server-side CRD schema validation from the CHART's CRDs — the same
+    schemas a real cluster installs — so controller writes a real
+    apiserver would 422 fail here too."""
+    import os
+
+    from ..kube.crdschema import CRDValidator
+
+    crds_dir = os.path.join(
+        os.path.dirname(os.path.dirname(os.path.dirname(os.path.abspath(__file__)))),
+        "charts",
+        "gpu-provisioner-amd",
+        "crds",
+    )
+    mapping = {
+        ("karpenter.sh/v1", "NodeClaim"): "karpenter.sh_nodeclaims.yaml",
+        ("kaito.sh/v1alpha1", "KaitoNodeClass"): "kaito.sh_kaitonodeclasses.yaml",
+    }
+    for gvk, fname in mapping.items():
+        path = os.path.join(crds_dir, fname)
+        if os.path.exists(path):
+            server.validators[gvk] = CRDValidator.from_file(path)
+
+
 class Harness:
     def __init__(
         self,
@@ -41,6 +65,7 @@ class Harness:
     ):
         self._gc_pacer_enabled = gc_pacer
         self.server = InMemoryAPIServer()
+        install_chart_crd_validators(self.server)
         self.kube = InMemoryClient(self.server)
         self.catalog = InstanceTypeProvider(region)
         self.agent_pools = FakeAgentPools(

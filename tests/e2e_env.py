@@ -367,8 +367,11 @@ class HttpInProcessEnv(E2EEnvironment):
         from gpu_provisioner_amd.providers.instance.provider import InstanceProvider
         from gpu_provisioner_amd.providers.instancetype.catalog import InstanceTypeProvider
 
+        from gpu_provisioner_amd.fake.harness import install_chart_crd_validators
+
         self._harness_kw = harness_kw  # latency knobs
         self.server = InMemoryAPIServer()
+        install_chart_crd_validators(self.server)
         self.rest = RESTServerHandle(self.server)
         self.actor_client = InMemoryClient(self.server)
         self.catalog = InstanceTypeProvider()

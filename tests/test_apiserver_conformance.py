@@ -259,3 +259,82 @@ def test_no_new_finalizers_on_terminating_object():
             await kube.get(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "fin1")
 
     run(main())
+
+
+# ---------------------------------------------------------------------------
+# Server-side CRD schema validation (real apiservers validate CR writes
+# against the installed CRD's structural schema + CEL; the harness server
+# carries the CHART's CRDs — fake/harness.install_chart_crd_validators)
+# ---------------------------------------------------------------------------
+
+
+def _valid_spec():
+    return {
+        "requirements": [
+            {"key": karpv1.INSTANCE_TYPE_LABEL_KEY, "operator": "In",
+             "values": ["Standard_ND128isr_MI355X_v6"]}
+        ],
+        "nodeClassRef": {"group": "kaito.sh", "kind": "KaitoNodeClass", "name": "d"},
+    }
+
+
+def test_crd_schema_enforced_on_create():
+    async def main():
+        h = Harness()
+        # missing required spec.requirements/nodeClassRef → 422
+        with pytest.raises(InvalidError, match="required"):
+            await h.kube.create(karpv1.new_nodeclaim("bad1", labels={}))
+        # invalid requirement operator enum → 422
+        nc = karpv1.new_nodeclaim("bad2", labels={})
+        nc["spec"] = _valid_spec()
+        nc["spec"]["requirements"][0]["operator"] = "Matches"
+        with pytest.raises(InvalidError, match="enum"):
+            await h.kube.create(nc)
+        # empty nodeClassRef.kind → CEL non-empty rule
+        nc = karpv1.new_nodeclaim("bad3", labels={})
+        nc["spec"] = _valid_spec()
+        nc["spec"]["nodeClassRef"]["kind"] = ""
+        with pytest.raises(InvalidError, match="empty"):
+            await h.kube.create(nc)
+        # a valid claim is admitted and gets the expireAfter DEFAULT applied
+        good = karpv1.new_nodeclaim("good1", labels={})
+        good["spec"] = _valid_spec()
+        created = await h.kube.create(good)
+        assert created["spec"]["expireAfter"] == "720h"
+
+    run(main())
+
+
+def test_crd_spec_immutability_cel_enforced():
+    async def main():
+        h = Harness()
+        nc = karpv1.new_nodeclaim("imm1", labels={})
+        nc["spec"] = _valid_spec()
+        created = await h.kube.create(nc)
+        created["spec"]["requirements"][0]["values"] = ["Standard_ND64is_MI355X_v6"]
+        with pytest.raises(InvalidError, match="immutable"):
+            await h.kube.update(created)
+        # metadata/status writes remain allowed
+        fresh = await h.kube.get(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "imm1")
+        fresh["metadata"]["labels"] = {"a": "b"}
+        await h.kube.update(fresh)
+        fresh = await h.kube.get(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "imm1")
+        fresh["status"] = {"providerID": "azure:///x"}
+        await h.kube.update_status(fresh)
+
+    run(main())
+
+
+def test_crd_quantity_pattern_enforced_on_status():
+    async def main():
+        h = Harness()
+        nc = karpv1.new_nodeclaim("qty1", labels={})
+        nc["spec"] = _valid_spec()
+        created = await h.kube.create(nc)
+        created["status"] = {"capacity": {"amd.com/gpu": "eight"}}  # not a quantity
+        with pytest.raises(InvalidError):
+            await h.kube.update_status(created)
+        created["status"] = {"capacity": {"amd.com/gpu": "8", "memory": "2048Gi"}}
+        await h.kube.update_status(created)
+
+    run(main())
