@@ -224,6 +224,11 @@ def main() -> None:
                                 f" + node-ready {args.ready_latency}s"
                             )
                         ),
+                        # round-2 apiserver realism: every write is admission-
+                        # validated against the chart's CRD schemas in-process
+                        # (r01 benches had no admission — numbers compare
+                        # accordingly)
+                        "apiserver": "in-memory, server-side CRD validation on",
                         "cycle": "create→Launched→Registered→Initialized(amd.com/gpu)→delete→gone",
                         "p50_ready_latency_s": round(p50, 4),
                         "p95_ready_latency_s": round(p95, 4),

@@ -286,6 +286,12 @@ class InMemoryAPIServer:
                     nm.pop(f, None)
             if new.get("spec") != cur.get("spec"):
                 nm["generation"] = cm.get("generation", 1) + 1
+            elif "spec" in new:
+                # re-share the unchanged spec subtree with the previous
+                # revision: deep_copy broke identity, and identity is what
+                # lets the CRD validator's changed-only walk (and any other
+                # revision differ) prune in O(1) instead of deep-comparing
+                new["spec"] = cur.get("spec")
             stored = new
         self._validate(gvk, stored, cur)  # CRD schema + CEL immutability (422)
         self._bump(stored)
