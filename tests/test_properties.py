@@ -266,3 +266,118 @@ def test_parse_duration_compositional(h, m, s):
         assert got is None  # k8s semantics: zero duration = unset here
     else:
         assert got is not None and got.total_seconds() == total
+
+
+# ---------------------------------------------------------------------------
+# CRD validator: changed-only pruning must be equivalent to full validation
+# ---------------------------------------------------------------------------
+
+import os
+
+import yaml as _yaml
+
+from gpu_provisioner_amd.kube.crdschema import CRDValidator, validate as crd_validate
+
+_CRD_PATH = os.path.join(
+    os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+    "charts", "gpu-provisioner-amd", "crds", "karpenter.sh_nodeclaims.yaml",
+)
+_VALIDATOR = CRDValidator.from_file(_CRD_PATH)
+
+
+@st.composite
+def valid_nodeclaim(draw):
+    name = draw(st.text("abcdefgh", min_size=1, max_size=8))
+    n_reqs = draw(st.integers(1, 3))
+    reqs = []
+    for i in range(n_reqs):
+        op = draw(st.sampled_from(["In", "NotIn", "Exists"]))
+        r = {"key": "node.kubernetes.io/instance-type" if i == 0 else f"amd.com/k{i}",
+             "operator": op}
+        if op != "Exists":
+            r["values"] = draw(
+                st.lists(st.text("abcXYZ019", min_size=1, max_size=8), min_size=1, max_size=3)
+            )
+        reqs.append(r)
+    nc = {
+        "apiVersion": "karpenter.sh/v1",
+        "kind": "NodeClaim",
+        "metadata": {"name": name},
+        "spec": {
+            "requirements": reqs,
+            "nodeClassRef": {"group": "kaito.sh", "kind": "KaitoNodeClass", "name": "d"},
+        },
+        "status": {
+            "providerID": draw(st.sampled_from(["", "azure:///x"])),
+            "capacity": {"amd.com/gpu": str(draw(st.integers(1, 8)))},
+            "conditions": [
+                {"type": "Launched", "status": draw(st.sampled_from(["True", "False"])),
+                 "reason": "Launched", "lastTransitionTime": "2026-01-01T00:00:00Z"}
+            ],
+        },
+    }
+    return nc
+
+
+@st.composite
+def mutation(draw):
+    """A mutation applied to a valid claim — possibly invalidating."""
+    kind = draw(st.sampled_from([
+        "cond_bad_status", "cond_add", "capacity_bad_qty", "capacity_good",
+        "req_bad_operator", "req_noop", "nodename", "too_many_reqs",
+    ]))
+    return kind
+
+
+def _apply_mutation(nc, kind):
+    import copy
+
+    out = copy.deepcopy(nc)
+    if kind == "cond_bad_status":
+        out["status"]["conditions"][0]["status"] = "Perhaps"  # not in enum
+    elif kind == "cond_add":
+        out["status"]["conditions"].append(
+            {"type": "Registered", "status": "True", "reason": "R",
+             "lastTransitionTime": "2026-01-01T00:00:01Z"}
+        )
+    elif kind == "capacity_bad_qty":
+        out["status"]["capacity"]["amd.com/gpu"] = "eight!"
+    elif kind == "capacity_good":
+        out["status"]["capacity"]["memory"] = "2048Gi"
+    elif kind == "req_bad_operator":
+        out["spec"]["requirements"][0]["operator"] = "Matches"
+    elif kind == "req_noop":
+        pass
+    elif kind == "nodename":
+        out["status"]["nodeName"] = "aks-x-123-vmss000000"
+    elif kind == "too_many_reqs":
+        out["spec"]["requirements"] = out["spec"]["requirements"] * 51  # > maxItems 100
+    return out
+
+
+@given(valid_nodeclaim(), mutation())
+@settings(max_examples=120, deadline=None)
+def test_changed_only_validation_equivalent_to_full(nc, kind):
+    """The validator's changed-subtree pruning (an optimization) must agree
+    with FULL schema validation on whether the new object is valid —
+    otherwise the fake admits writes a real apiserver would reject."""
+    new = _apply_mutation(nc, kind)
+    full_errs = crd_validate(new, _VALIDATOR.schema)
+    pruned_errs = _VALIDATOR(new, nc)  # update path: diffs against old
+    # immutability: spec mutations are CEL-rejected on update regardless of
+    # structural validity — exclude that extra error from the comparison
+    pruned_structural = [e for e in pruned_errs if "immutable" not in e]
+    assert bool(full_errs) == bool(pruned_structural), (
+        f"divergence for {kind}: full={full_errs} pruned={pruned_structural}"
+    )
+    # and the create path must agree with full validation too
+    create_errs = _VALIDATOR(copy_for_create(new), None)
+    assert bool(full_errs) == bool(create_errs), (
+        f"create divergence for {kind}: full={full_errs} create={create_errs}"
+    )
+
+
+def copy_for_create(obj):
+    import copy
+
+    return copy.deepcopy(obj)
