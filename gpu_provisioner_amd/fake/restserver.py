@@ -103,6 +103,13 @@ def build_app(server: InMemoryAPIServer) -> Starlette:
         subresource = request.path_params.get("subresource", "")
         q = request.query_params
         try:
+            if request.method == "GET" and subresource == "log" and kind == "Pod":
+                # pod log subresource: real content comes from the kubelet;
+                # here a canned line is enough for the diagnostics path
+                await server.get(api_version, kind, name, namespace)  # 404 check
+                return Response(
+                    f"(fake log) pod {namespace}/{name}\n", media_type="text/plain"
+                )
             if request.method == "GET" and not name:
                 if q.get("watch") in ("true", "1"):
                     return await watch_stream(api_version, kind, q)
@@ -199,19 +206,23 @@ def build_app(server: InMemoryAPIServer) -> Starlette:
 
         return StreamingResponse(gen(), media_type="application/json")
 
+    # literal `namespaces/` patterns MUST precede the generic ones —
+    # starlette matches in registration order and
+    # /api/v1/namespaces/<ns>/pods would otherwise bind to
+    # {resource}/{name}/{subresource}
     patterns = [
-        "/api/{version}/{resource}",
-        "/api/{version}/{resource}/{name}",
-        "/api/{version}/{resource}/{name}/{subresource}",
         "/api/{version}/namespaces/{namespace}/{resource}",
         "/api/{version}/namespaces/{namespace}/{resource}/{name}",
         "/api/{version}/namespaces/{namespace}/{resource}/{name}/{subresource}",
-        "/apis/{group}/{version}/{resource}",
-        "/apis/{group}/{version}/{resource}/{name}",
-        "/apis/{group}/{version}/{resource}/{name}/{subresource}",
+        "/api/{version}/{resource}",
+        "/api/{version}/{resource}/{name}",
+        "/api/{version}/{resource}/{name}/{subresource}",
         "/apis/{group}/{version}/namespaces/{namespace}/{resource}",
         "/apis/{group}/{version}/namespaces/{namespace}/{resource}/{name}",
         "/apis/{group}/{version}/namespaces/{namespace}/{resource}/{name}/{subresource}",
+        "/apis/{group}/{version}/{resource}",
+        "/apis/{group}/{version}/{resource}/{name}",
+        "/apis/{group}/{version}/{resource}/{name}/{subresource}",
     ]
     methods = ["GET", "POST", "PUT", "PATCH", "DELETE"]
     routes = [Route(p, handle, methods=methods) for p in patterns]
