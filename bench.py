@@ -195,6 +195,11 @@ def main() -> None:
         if len(latencies) >= 20
         else (max(latencies) if latencies else float("nan"))
     )
+    p99 = (
+        statistics.quantiles(latencies, n=100)[98]
+        if len(latencies) >= 100
+        else (max(latencies) if latencies else float("nan"))
+    )
 
     if rank == 0:
         print(
@@ -232,6 +237,7 @@ def main() -> None:
                         "cycle": "create→Launched→Registered→Initialized(amd.com/gpu)→delete→gone",
                         "p50_ready_latency_s": round(p50, 4),
                         "p95_ready_latency_s": round(p95, 4),
+                        "p99_ready_latency_s": round(p99, 4),
                         "parallelism": f"dp{world_size}" if world_size > 1 else "single",
                     },
                 }
