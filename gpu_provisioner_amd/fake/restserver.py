@@ -172,6 +172,13 @@ def build_app(server: InMemoryAPIServer) -> Starlette:
         return JSONResponse(_status_body(405, "MethodNotAllowed", request.method), status_code=405)
 
     async def watch_stream(api_version: str, kind: str, q) -> Response:
+        from ..kube.client import LabelSelector
+
+        sel = (
+            LabelSelector.parse(q.get("labelSelector", ""))
+            if q.get("labelSelector")
+            else None
+        )
         rv = q.get("resourceVersion", "")
         try:
             queue, unsubscribe = server.subscribe(api_version, kind, rv)
@@ -200,6 +207,10 @@ def build_app(server: InMemoryAPIServer) -> Starlette:
                         return
                     if obj is None:  # _WATCH_BROKEN chaos sentinel
                         return
+                    if sel is not None and not sel.matches(
+                        (obj.get("metadata") or {}).get("labels") or {}
+                    ):
+                        continue
                     yield json.dumps({"type": event_type, "object": obj}) + "\n"
             finally:
                 unsubscribe()
