@@ -40,6 +40,13 @@ log = logging.getLogger(__name__)
 _THRESHOLDS = (700, 10, 1_000_000_000)
 DEFAULT_GEN2_INTERVAL = 10.0
 
+# process-wide engagement count: with multiple managers in one process
+# (dual-replica HA tests), only the first engage freezes/retunes and only
+# the last disengage restores — otherwise the first stop() would re-enable
+# automatic gen2 under the survivor
+_active = 0
+_saved_thresholds: Optional[tuple] = None
+
 
 class GCPacer:
     def __init__(self, gen2_interval: float = DEFAULT_GEN2_INTERVAL, freeze: bool = True):
@@ -52,13 +59,16 @@ class GCPacer:
     def engage(self) -> None:
         """Call once the process reached steady state (informers synced,
         controllers constructed): the current heap is what gets frozen."""
+        global _active, _saved_thresholds
         if self._engaged:
             return
-        self._saved_thresholds = gc.get_threshold()
-        gc.collect()  # the frozen set must hold no cycle garbage
-        if self.freeze:
-            gc.freeze()
-        gc.set_threshold(*_THRESHOLDS)
+        if _active == 0:
+            _saved_thresholds = gc.get_threshold()
+            gc.collect()  # the frozen set must hold no cycle garbage
+            if self.freeze:
+                gc.freeze()
+            gc.set_threshold(*_THRESHOLDS)
+        _active += 1
         self._engaged = True
         self._task = asyncio.create_task(self._run(), name="gc-pacer")
 
@@ -78,6 +88,7 @@ class GCPacer:
                 )
 
     async def disengage(self) -> None:
+        global _active, _saved_thresholds
         if not self._engaged:
             return
         if self._task is not None:
@@ -87,8 +98,11 @@ class GCPacer:
             except (asyncio.CancelledError, Exception):
                 pass
             self._task = None
-        if self.freeze:
-            gc.unfreeze()
-        if self._saved_thresholds is not None:
-            gc.set_threshold(*self._saved_thresholds)
         self._engaged = False
+        _active -= 1
+        if _active == 0:
+            if self.freeze:
+                gc.unfreeze()
+            if _saved_thresholds is not None:
+                gc.set_threshold(*_saved_thresholds)
+                _saved_thresholds = None

@@ -37,3 +37,20 @@ def test_pacer_engage_idempotent():
         await p.disengage()
 
     run(main())
+
+
+def test_nested_pacers_restore_only_at_last_disengage():
+    """Two managers in one process (HA tests): the first disengage must
+    NOT re-enable automatic gen2 under the survivor."""
+
+    async def main():
+        before = gc.get_threshold()
+        p1, p2 = GCPacer(gen2_interval=60.0), GCPacer(gen2_interval=60.0)
+        p1.engage()
+        p2.engage()
+        await p1.disengage()
+        assert gc.get_threshold()[2] >= 1_000_000  # p2 still paced
+        await p2.disengage()
+        assert gc.get_threshold() == before
+
+    run(main())

@@ -9,7 +9,6 @@ an untested claim. Here every line of it runs against the in-memory
 apiserver over real HTTP, with the controller manager in-process and the
 AKS simulator as the cloud/kubelet actor."""
 import asyncio
-import os
 
 import pytest
 import yaml
