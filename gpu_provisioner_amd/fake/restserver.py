@@ -197,6 +197,12 @@ def build_app(server: InMemoryAPIServer) -> Starlette:
             return StreamingResponse(gone_gen(), media_type="application/json")
 
         async def gen():
+            # real apiservers interleave BOOKMARK events (allowWatchBookmarks)
+            # so clients can advance their resume rv without object traffic;
+            # emit one every N delivered events
+            bookmark_every = 25
+            since_bookmark = 0
+            last_rv = rv
             try:
                 while True:
                     try:
@@ -207,11 +213,25 @@ def build_app(server: InMemoryAPIServer) -> Starlette:
                         return
                     if obj is None:  # _WATCH_BROKEN chaos sentinel
                         return
+                    last_rv = (obj.get("metadata") or {}).get("resourceVersion", last_rv)
                     if sel is not None and not sel.matches(
                         (obj.get("metadata") or {}).get("labels") or {}
                     ):
                         continue
                     yield json.dumps({"type": event_type, "object": obj}) + "\n"
+                    since_bookmark += 1
+                    if since_bookmark >= bookmark_every:
+                        since_bookmark = 0
+                        yield json.dumps(
+                            {
+                                "type": "BOOKMARK",
+                                "object": {
+                                    "kind": kind,
+                                    "apiVersion": api_version,
+                                    "metadata": {"resourceVersion": last_rv},
+                                },
+                            }
+                        ) + "\n"
             finally:
                 unsubscribe()
 

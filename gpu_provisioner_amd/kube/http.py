@@ -388,6 +388,11 @@ class HTTPClient(KubeClient):
                         raise GoneError(obj.get("message", "watch expired"))
                     raise APIError(obj.get("message", "watch error"))
                 if etype == "BOOKMARK":
+                    # surface bookmarks: the informer advances its resume
+                    # resourceVersion from them (client-go reflector
+                    # behavior), which keeps relists cheap after long quiet
+                    # periods
+                    yield etype, obj
                     continue
                 obj.setdefault("apiVersion", api_version)
                 obj.setdefault("kind", kind)

@@ -14,7 +14,7 @@ from collections import defaultdict
 from typing import Callable, Optional
 
 from . import objects as ko
-from .client import ADDED, DELETED, GoneError, KubeClient, MODIFIED
+from .client import ADDED, BOOKMARK, DELETED, GoneError, KubeClient, MODIFIED
 
 log = logging.getLogger(__name__)
 
@@ -285,6 +285,8 @@ class Informer:
             self.api_version, self.kind, self.namespace, self._rv, self.label_selector
         ):
             self._rv = obj.get("metadata", {}).get("resourceVersion", self._rv)
+            if event_type == BOOKMARK:
+                continue  # rv advanced above; no object state change
             if event_type == DELETED:
                 key = object_key(obj)
                 old = self._cache.pop(key, None)
