@@ -111,8 +111,10 @@ def test_watch_stream_decoding_and_gone():
         got = []
         with pytest.raises(GoneError):
             async for etype, obj in c.watch("v1", "Node", resource_version="7"):
-                got.append((etype, obj["metadata"]["name"]))
-        assert got == [("ADDED", "a"), ("MODIFIED", "a")]  # bookmark skipped
+                got.append((etype, obj["metadata"].get("resourceVersion")))
+        # bookmarks are SURFACED (the informer advances its resume rv from
+        # them) — they are not object events, just rv carriers
+        assert got == [("ADDED", "1"), ("BOOKMARK", "2"), ("MODIFIED", "3")]
 
     run(main())
 
