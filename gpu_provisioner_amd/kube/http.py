@@ -127,6 +127,7 @@ class HTTPClient(KubeClient):
         path = path or os.environ.get("KUBECONFIG", os.path.expanduser("~/.kube/config"))
         with open(path) as f:
             cfg = yaml.safe_load(f)
+        tmp_files: list = []
 
         def by_name(section: str, name: str) -> dict:
             for entry in cfg.get(section, []):
@@ -153,6 +154,7 @@ class HTTPClient(KubeClient):
                 tmp = tempfile.NamedTemporaryFile(delete=False, suffix=".pem")
                 tmp.write(base64.b64decode(user[data_key]))
                 tmp.close()
+                tmp_files.append(tmp.name)
                 return tmp.name
             return ""
 
@@ -169,6 +171,7 @@ class HTTPClient(KubeClient):
             ca_tmp = tempfile.NamedTemporaryFile(delete=False, suffix=".crt")
             ca_tmp.write(base64.b64decode(cluster["certificate-authority-data"]))
             ca_tmp.close()
+            tmp_files.append(ca_tmp.name)
             verify = ca_tmp.name
         cert_file = materialize("client-certificate-data", "client-certificate")
         key_file = materialize("client-key-data", "client-key")
@@ -181,9 +184,12 @@ class HTTPClient(KubeClient):
                 sslctx.verify_mode = ssl.CERT_NONE
             sslctx.load_cert_chain(cert_file, key_file)
             verify = sslctx
-        return cls(
+        client = cls(
             cluster["server"], token=token, verify=verify, qps=qps, burst=burst
         )
+        # materialized cert/CA files are cleaned up with the client
+        client._tmp_files = tmp_files
+        return client
 
     # ------------------------------------------------------------ plumbing
 
@@ -438,3 +444,8 @@ class HTTPClient(KubeClient):
 
     async def close(self) -> None:
         await self.http.aclose()
+        for p in getattr(self, "_tmp_files", ()):  # from_kubeconfig materializations
+            try:
+                os.unlink(p)
+            except OSError:
+                pass
