@@ -231,3 +231,19 @@ def test_catalog_sku_file_override(tmp_path):
     bad.write_text(yaml.safe_dump([{"name": "X", "vcpu": 1}]))
     with pytest.raises(ValueError, match="missing fields"):
         InstanceTypeProvider(sku_file=str(bad))
+
+
+def test_shipped_sku_override_example_loads():
+    """examples/sku-catalog-override.yaml must keep loading through the
+    real GPU_PROV_SKU_FILE code path (price override, next-gen add,
+    removal)."""
+    import os
+
+    path = os.path.join(
+        os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+        "examples", "sku-catalog-override.yaml",
+    )
+    cat = InstanceTypeProvider(sku_file=path)
+    assert cat.get("Standard_ND96isr_MI400X_v7") is not None
+    assert cat.gpu_count("Standard_ND96isr_MI400X_v7") == 8
+    assert cat.get("Standard_ND32is_MI355X_v6") is None
