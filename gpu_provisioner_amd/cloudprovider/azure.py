@@ -18,7 +18,7 @@ from ..kube import objects as ko
 from ..providers.instance import bootstrap
 from ..providers.instance.provider import InstanceProvider
 from ..providers.instancetype.catalog import InstanceTypeProvider
-from .types import CloudProvider, Instance, NodeClaimNotFoundError, RepairPolicy
+from .types import CloudProvider, Instance, RepairPolicy
 
 NODE_REPAIR_TOLERATION_SECONDS = 600.0  # 10 min
 # GPU-sick nodes (nodeagent's AMDGPUHealthy=False: failed HBM/MFMA/LDS/xGMI
@@ -44,6 +44,13 @@ class AzureCloudProvider(CloudProvider):
     ):
         self.instances = instances
         self.catalog = catalog
+        # drift-sweep instance view: ONE paged agent-pool LIST per sweep
+        # instead of an ARM GET per claim — a 10k-claim fleet would
+        # otherwise issue ~83 GETs/s against ARM every 2 min. The drift
+        # controller invalidates at the start of each sweep; the TTL is a
+        # staleness backstop for ad-hoc is_drifted calls.
+        self._drift_cache: tuple = (0.0, None)  # (fetched_at_monotonic, {pool: Instance})
+        self.drift_cache_ttl = 30.0
         self._repair_policies = [
             RepairPolicy("Ready", ko.CONDITION_FALSE, repair_toleration),
             RepairPolicy("Ready", ko.CONDITION_UNKNOWN, repair_toleration),
@@ -89,10 +96,9 @@ class AzureCloudProvider(CloudProvider):
         pid = karpv1.provider_id_of(nodeclaim)
         if not pid:
             return ""
-        try:
-            instance = await self.instances.get(pid)
-        except NodeClaimNotFoundError:
-            return ""
+        instance = await self._instance_for_drift(nodeclaim, pid)
+        if instance is None:
+            return ""  # vanished instance is the GC controllers' domain
         wanted = karpv1.requirement_values(nodeclaim, karpv1.INSTANCE_TYPE_LABEL_KEY)
         if wanted and instance.type and instance.type not in wanted:
             return DRIFT_INSTANCE_TYPE
@@ -120,6 +126,27 @@ class AzureCloudProvider(CloudProvider):
                 if req.key in labels and not req.has(labels[req.key]):
                     return DRIFT_REQUIREMENTS
         return ""
+
+    def invalidate_drift_cache(self) -> None:
+        """Called by the drift controller at the start of every sweep so
+        each sweep judges a fresh cloud snapshot."""
+        self._drift_cache = (0.0, None)
+
+    async def _instance_for_drift(self, nodeclaim: dict, provider_id: str):
+        """Instance view for the drift sweep, served from a short-TTL
+        snapshot of the paged agent-pool LIST (one ARM call per window,
+        O(1) per claim)."""
+        import time as _time
+
+        from ..utils.utils import parse_agent_pool_name_from_id
+
+        now = _time.monotonic()
+        fetched_at, cache = self._drift_cache
+        if cache is None or now - fetched_at > self.drift_cache_ttl:
+            cache = {i.name: i for i in await self.instances.list()}
+            self._drift_cache = (now, cache)
+        pool = parse_agent_pool_name_from_id(provider_id) or ko.name_of(nodeclaim)
+        return cache.get(pool)
 
     async def _node_for(self, provider_id: str):
         inf = getattr(self.instances, "nodes_informer", None)

@@ -26,6 +26,11 @@ class MetricsDecorator(CloudProvider):
     def __init__(self, inner: CloudProvider):
         self.inner = inner
 
+    def __getattr__(self, name: str):
+        # forward provider-specific extensions (e.g. invalidate_drift_cache)
+        # that are outside the metrics-decorated CloudProvider contract
+        return getattr(self.inner, name)
+
     def _observe(self, method: str, start: float, err: Optional[BaseException]) -> None:
         controller = current_controller.get()
         CLOUDPROVIDER_DURATION.labels(

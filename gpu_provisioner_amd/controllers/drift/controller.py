@@ -57,6 +57,11 @@ class DriftController:
 
     async def reconcile(self, key: str) -> Optional[Result]:
         decorator.current_controller.set(self.NAME)
+        # fresh cloud snapshot per sweep (one paged LIST; the provider
+        # serves every per-claim is_drifted from it)
+        invalidate = getattr(self.cloud, "invalidate_drift_cache", None)
+        if invalidate is not None:
+            invalidate()
         claims = await self.kube.list(karpv1.API_VERSION, karpv1.KIND_NODECLAIM)
         candidates = [
             nc
