@@ -28,7 +28,11 @@ class MetricsDecorator(CloudProvider):
 
     def __getattr__(self, name: str):
         # forward provider-specific extensions (e.g. invalidate_drift_cache)
-        # that are outside the metrics-decorated CloudProvider contract
+        # that are outside the metrics-decorated CloudProvider contract.
+        # Guard 'inner' itself: before __init__ assigns it (copy/unpickle
+        # paths) forwarding would recurse infinitely.
+        if name == "inner":
+            raise AttributeError(name)
         return getattr(self.inner, name)
 
     def _observe(self, method: str, start: float, err: Optional[BaseException]) -> None:
