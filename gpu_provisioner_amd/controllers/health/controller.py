@@ -65,15 +65,27 @@ class HealthController:
             for k in [k for k in self._first_seen if k[0] == name]:
                 del self._first_seen[k]
             return
-        if karpv1.node_is_managed(obj):
-            self.controller.enqueue_nowait(ko.name_of(obj))
+        if not karpv1.node_is_managed(obj):
+            return
+        # fleet-scale pre-filter: kubelet heartbeats dominate node events;
+        # a node with no policy-matching condition needs no reconcile (the
+        # toleration timer rides on the reconcile's own requeue). A node
+        # that LATER degrades fires this handler again with the unhealthy
+        # condition present.
+        if self._match_policy(obj) is None:
+            return
+        self.controller.enqueue_nowait(ko.name_of(obj))
 
     async def reconcile(self, key: str) -> Optional[Result]:
         decorator.current_controller.set(self.NAME)
-        try:
-            node = await self.kube.get("v1", "Node", key)
-        except NotFoundError:
-            return None
+        # read-only reconcile: the informer cache suffices (no mutation of
+        # the node object below) — a GET per heartbeat is O(cluster) load
+        node = self.nodes.get(key) if self.nodes.has_synced else None
+        if node is None:
+            try:
+                node = await self.kube.get("v1", "Node", key)
+            except NotFoundError:
+                return None
         if not karpv1.node_is_managed(node) or ko.is_deleting(node):
             return None
         match = self._match_policy(node)
