@@ -338,3 +338,27 @@ def test_crd_quantity_pattern_enforced_on_status():
         await h.kube.update_status(created)
 
     run(main())
+
+
+def test_delete_uid_precondition_conflicts():
+    """DeleteOptions.preconditions.uid mismatch → 409 Conflict (API
+    conventions; the GC/drift/repair deletes all carry the observed uid so
+    a name-reuse race can never delete the replacement object)."""
+
+    async def main():
+        kube = client()
+        nc = await kube.create(karpv1.new_nodeclaim("uid1", labels={}))
+        with pytest.raises(ConflictError):
+            await kube.delete(
+                karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "uid1",
+                uid_precondition="some-other-uid",
+            )
+        # matching uid deletes
+        await kube.delete(
+            karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "uid1",
+            uid_precondition=nc["metadata"]["uid"],
+        )
+        with pytest.raises(Exception):
+            await kube.get(karpv1.API_VERSION, karpv1.KIND_NODECLAIM, "uid1")
+
+    run(main())
