@@ -168,7 +168,12 @@ def build_app(server: InMemoryAPIServer) -> Starlette:
                 return JSONResponse(_status_body(200, "", "deleted") | {"status": "Success"})
         except APIError as e:
             code, reason = _status_for(e)
-            return JSONResponse(_status_body(code, reason, str(e)), status_code=code)
+            headers = {}
+            if isinstance(e, TooManyRequestsError):
+                headers["Retry-After"] = str(int(getattr(e, "retry_after", 1) or 1))
+            return JSONResponse(
+                _status_body(code, reason, str(e)), status_code=code, headers=headers
+            )
         return JSONResponse(_status_body(405, "MethodNotAllowed", request.method), status_code=405)
 
     async def watch_stream(api_version: str, kind: str, q) -> Response:
