@@ -386,7 +386,6 @@ class HttpInProcessEnv(E2EEnvironment):
             plugin_latency=harness_kw.get("plugin_latency", 0.05),
             gpu_count_for=self.catalog.gpu_count,
         )
-        self._azure_cloud = None
         self.manager = None
         self.run_id = f"http-{uuid.uuid4().hex[:8]}"
         self.default_timeout = 30.0
