@@ -22,6 +22,12 @@ LEASE_DURATION = 15.0
 RENEW_INTERVAL = 5.0
 RETRY_INTERVAL = 2.0
 
+# renewal outcomes: transient errors must NOT yield leadership while the
+# lease is still live (client-go retries until its renewDeadline)
+RENEW_OK = "ok"
+RENEW_LOST = "lost"  # another holder took the lease — yield immediately
+RENEW_ERROR = "error"  # transient API failure — retry until the deadline
+
 
 class LeaderElector:
     def __init__(
@@ -32,6 +38,7 @@ class LeaderElector:
         identity: str = "",
         lease_duration: float = LEASE_DURATION,
         renew_interval: float = RENEW_INTERVAL,
+        renew_deadline: Optional[float] = None,
     ):
         self.kube = kube
         self.name = name
@@ -39,6 +46,12 @@ class LeaderElector:
         self.identity = identity or f"{socket.gethostname()}_{uuid.uuid4().hex[:8]}"
         self.lease_duration = lease_duration
         self.renew_interval = renew_interval
+        # how long renewals may keep FAILING before leadership is yielded
+        # (client-go's renewDeadline shape: under the lease duration so we
+        # stop acting before a rival can acquire the expired lease)
+        self.renew_deadline = (
+            renew_deadline if renew_deadline is not None else lease_duration * 2.0 / 3.0
+        )
         self.is_leader = False
         self._task: Optional[asyncio.Task] = None
 
@@ -54,9 +67,25 @@ class LeaderElector:
             log.info("leader election: acquired lease %s/%s", self.namespace, self.name)
             await on_started_leading()
             try:
+                import time as _time
+
+                last_ok = _time.monotonic()
                 while True:
                     await asyncio.sleep(self.renew_interval)
-                    if not await self._renew():
+                    outcome = await self._renew()
+                    if outcome == RENEW_OK:
+                        last_ok = _time.monotonic()
+                        continue
+                    if outcome == RENEW_LOST:
+                        break  # another holder: yield immediately
+                    # transient failure: keep leading and retrying until the
+                    # renew deadline — a single API blip must not stop the
+                    # fleet's only active controller
+                    if _time.monotonic() - last_ok > self.renew_deadline:
+                        log.warning(
+                            "leader election: renewals failing for %.0fs (deadline %.0fs)",
+                            _time.monotonic() - last_ok, self.renew_deadline,
+                        )
                         break
             finally:
                 self.is_leader = False
@@ -125,16 +154,21 @@ class LeaderElector:
         except Exception as e:
             log.warning("leader election: lease release failed (expiry will cover): %s", e)
 
-    async def _renew(self) -> bool:
+    async def _renew(self) -> str:
         try:
             cur = await self.kube.get("coordination.k8s.io/v1", "Lease", self.name, self.namespace)
+        except NotFoundError:
+            return RENEW_LOST  # lease deleted out from under us
         except Exception:
-            return False
+            return RENEW_ERROR
         if cur.get("spec", {}).get("holderIdentity") != self.identity:
-            return False
+            return RENEW_LOST
         cur["spec"]["renewTime"] = ko.fmt_micro_time(ko.now())
         try:
             await self.kube.update(cur)
-            return True
+            return RENEW_OK
+        except (ConflictError, NotFoundError):
+            # concurrent write to OUR lease — re-judge ownership next tick
+            return RENEW_ERROR
         except Exception:
-            return False
+            return RENEW_ERROR

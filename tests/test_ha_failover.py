@@ -135,3 +135,89 @@ def test_graceful_stop_releases_lease_for_fast_takeover():
             await m2.stop()
 
     run(main(), timeout=120)
+
+
+def test_transient_renew_errors_do_not_yield_leadership():
+    """client-go renewDeadline semantics: a renewal failing on transient
+    API errors keeps leading until the deadline; a sustained outage or a
+    stolen lease yields."""
+    from gpu_provisioner_amd.kube.client import APIError
+    from gpu_provisioner_amd.operator.leaderelection import LeaderElector
+
+    async def main():
+        h = Harness(gc_pacer=False)
+        el = LeaderElector(
+            h.kube, "lease1", "kube-system",
+            lease_duration=30.0, renew_interval=0.05, renew_deadline=0.6,
+        )
+        started, stopped = asyncio.Event(), asyncio.Event()
+
+        async def on_start(): started.set()
+        async def on_stop(): stopped.set()
+
+        task = asyncio.create_task(el.run(on_start, on_stop))
+        try:
+            await asyncio.wait_for(started.wait(), 5)
+            assert el.is_leader
+
+            # inject transient Lease API failures shorter than the deadline
+            fail = {"on": True}
+
+            def reactor(verb, gvk, payload):
+                if gvk == ("coordination.k8s.io/v1", "Lease") and fail["on"]:
+                    return APIError("transient apiserver blip")
+                return None
+
+            h.server.reactors.append(reactor)
+            await asyncio.sleep(0.3)  # several failed renew ticks < deadline
+            assert el.is_leader, "transient renew errors must not yield leadership"
+            fail["on"] = False
+            await asyncio.sleep(0.2)
+            assert el.is_leader
+
+            # sustained outage past the deadline yields
+            fail["on"] = True
+            await asyncio.wait_for(stopped.wait(), 5)
+            assert not el.is_leader
+        finally:
+            task.cancel()
+            try:
+                await task
+            except (asyncio.CancelledError, Exception):
+                pass
+            h.server.reactors.clear()
+
+    run(main(), timeout=60)
+
+
+def test_stolen_lease_yields_immediately():
+    from gpu_provisioner_amd.operator.leaderelection import LeaderElector
+
+    async def main():
+        h = Harness(gc_pacer=False)
+        el = LeaderElector(
+            h.kube, "lease2", "kube-system",
+            lease_duration=30.0, renew_interval=0.05, renew_deadline=10.0,
+        )
+        started, stopped = asyncio.Event(), asyncio.Event()
+
+        async def on_start(): started.set()
+        async def on_stop(): stopped.set()
+
+        task = asyncio.create_task(el.run(on_start, on_stop))
+        try:
+            await asyncio.wait_for(started.wait(), 5)
+            # a rival takes the lease out-of-band
+            cur = await h.kube.get("coordination.k8s.io/v1", "Lease", "lease2", "kube-system")
+            cur["spec"]["holderIdentity"] = "rival"
+            await h.kube.update(cur)
+            await asyncio.wait_for(stopped.wait(), 5)
+            assert not el.is_leader
+        finally:
+            task.cancel()
+            try:
+                await task
+            except (asyncio.CancelledError, Exception):
+                pass
+
+    run(main(), timeout=60)
