@@ -184,10 +184,10 @@ def test_leader_election_single_holder_and_failover():
                            lease_duration=1.0, renew_interval=0.1)
         assert await e1._try_acquire() is True
         assert await e2._try_acquire() is False  # held and fresh
-        assert await e1._renew() is True
+        assert await e1._renew() == "ok"
         # holder dies; after lease_duration the lock is stealable
         await asyncio.sleep(1.2)
         assert await e2._try_acquire() is True
-        assert await e1._renew() is False  # a lost the lease
+        assert await e1._renew() == "lost"  # a lost the lease
 
     run(main())
