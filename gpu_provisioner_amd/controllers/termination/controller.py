@@ -299,36 +299,50 @@ class TerminationController:
     # --------------------------------------------------------------- helpers
 
     async def _taint(self, node: dict) -> None:
-        taints = ko.node_taints(node)
         desired = {"key": karpv1.DISRUPTED_TAINT_KEY, "effect": "NoSchedule"}
-        labels = ko.labels_of(node)
-        needs_taint = not any(
-            t.get("key") == desired["key"] and t.get("effect") == desired["effect"]
-            for t in taints
-        )
-        needs_label = labels.get(karpv1.EXCLUDE_FROM_LB_LABEL_KEY) != "karpenter"
-        if not needs_taint and not needs_label:
-            return
-        # optimistic lock on the taint merge: same lost-update race as
-        # registration's taint sync (ConflictError surfaces to reconcile,
-        # which requeues against a fresh read)
-        patch: dict = {
-            "metadata": {"resourceVersion": ko.meta(node).get("resourceVersion")},
-            "spec": {},
-        }
-        if needs_taint:
-            patch["spec"]["taints"] = ko.merge_taints(taints, [desired])
-        if needs_label:
-            patch["metadata"]["labels"] = {
-                **labels,
-                karpv1.EXCLUDE_FROM_LB_LABEL_KEY: "karpenter",
+        # optimistic lock on the taint merge (same lost-update race as
+        # registration's taint sync) with IN-PLACE conflict retries against
+        # a fresh read: surfacing every conflict as a rate-limited requeue
+        # funneled fleet-scale teardowns through the controller's token
+        # bucket (32k-claim storms collapsed to bucket speed)
+        for attempt in range(5):
+            taints = ko.node_taints(node)
+            labels = ko.labels_of(node)
+            needs_taint = not any(
+                t.get("key") == desired["key"] and t.get("effect") == desired["effect"]
+                for t in taints
+            )
+            needs_label = labels.get(karpv1.EXCLUDE_FROM_LB_LABEL_KEY) != "karpenter"
+            if not needs_taint and not needs_label:
+                return
+            patch: dict = {
+                "metadata": {"resourceVersion": ko.meta(node).get("resourceVersion")},
+                "spec": {},
             }
-        updated = await self.kube.patch("v1", "Node", ko.name_of(node), patch)
-        self._record_write(ko.name_of(node), updated)
-        # sync the in-hand node so later writes (finalizer removal) carry the
-        # post-taint rv instead of conflicting against our own patch
-        node.clear()
-        node.update(updated)
+            if needs_taint:
+                patch["spec"]["taints"] = ko.merge_taints(taints, [desired])
+            if needs_label:
+                patch["metadata"]["labels"] = {
+                    **labels,
+                    karpv1.EXCLUDE_FROM_LB_LABEL_KEY: "karpenter",
+                }
+            try:
+                updated = await self.kube.patch("v1", "Node", ko.name_of(node), patch)
+            except ConflictError:
+                try:
+                    fresh = await self.kube.get("v1", "Node", ko.name_of(node))
+                except NotFoundError:
+                    return  # node vanished mid-taint: nothing left to mark
+                node.clear()
+                node.update(fresh)
+                continue
+            self._record_write(ko.name_of(node), updated)
+            # sync the in-hand node so later writes (finalizer removal) carry
+            # the post-taint rv instead of conflicting against our own patch
+            node.clear()
+            node.update(updated)
+            return
+        raise ConflictError(f"node {ko.name_of(node)}: taint contention persists")
 
     async def _nodeclaim_for_node(self, node: dict) -> Optional[dict]:
         pid = ko.provider_id_of(node)

@@ -66,9 +66,13 @@ class TokenBucket:
 
 class RateLimiter:
     """Max of exponential per-item backoff and the global bucket (client-go's
-    DefaultControllerRateLimiter shape)."""
+    DefaultControllerRateLimiter shape). The bucket defaults are wider than
+    client-go's 10 qps/100: this provisioner runs hundreds-to-thousands of
+    concurrent reconciles, and a fleet-scale burst of conflict-requeues at
+    10/s collapsed 32k-claim teardowns to bucket speed; the per-item
+    exponential backoff still bounds hot error loops."""
 
-    def __init__(self, base: float = 0.005, cap: float = 1000.0, qps: float = 10.0, burst: int = 100):
+    def __init__(self, base: float = 0.005, cap: float = 1000.0, qps: float = 100.0, burst: int = 1000):
         self.backoff = ExponentialBackoff(base, cap)
         self.bucket = TokenBucket(qps, burst)
 
@@ -96,6 +100,12 @@ class RateLimitingQueue:
         self._getters: deque = deque()  # parked get() futures
         self._shutdown = False
         self.adds = 0  # metric: total adds
+        # ONE shared timer for the earliest delayed deadline. Per-getter
+        # timers put every idle worker's timer on the same instant — 1024
+        # workers thundering awake per delayed wave measurably saturated
+        # the loop during fleet-scale teardowns.
+        self._timer = None  # Optional[asyncio.TimerHandle]
+        self._timer_deadline = float("inf")
 
     def _wake_one(self) -> None:
         while self._getters:
@@ -137,10 +147,42 @@ class RateLimitingQueue:
             return
         self._seq += 1
         heapq.heappush(self._delayed, (time.monotonic() + delay, self._seq, item))
-        self._wake_one()  # a sleeping getter must recompute its timeout
+        # (re)arm the shared timer if this deadline is now the earliest —
+        # no getter wakes until an item is actually due
+        self._rearm_timer()
 
     async def add_rate_limited(self, item: Hashable) -> None:
         await self.add_after(item, self.rate_limiter.when(item))
+
+    def _rearm_timer(self) -> None:
+        """Keep exactly one timer armed at the earliest delayed deadline."""
+        if not self._delayed:
+            if self._timer is not None:
+                self._timer.cancel()
+                self._timer = None
+                self._timer_deadline = float("inf")
+            return
+        deadline = self._delayed[0][0]
+        if self._timer is not None:
+            if self._timer_deadline <= deadline + 1e-4:
+                return  # armed at-or-before the earliest deadline already
+            self._timer.cancel()
+        try:
+            loop = asyncio.get_running_loop()
+        except RuntimeError:
+            return  # outside the loop (constructor paths); get() re-arms
+        self._timer_deadline = deadline
+        self._timer = loop.call_later(
+            max(0.0, deadline - time.monotonic()), self._on_timer
+        )
+
+    def _on_timer(self) -> None:
+        self._timer = None
+        self._timer_deadline = float("inf")
+        # one getter wakes, drains the due items, takes one and chains a
+        # wakeup if more became ready (see get())
+        self._wake_one()
+        self._rearm_timer()
 
     async def get(self) -> Any:
         """Block until an item is ready; marks it processing. Returns None on shutdown."""
@@ -151,20 +193,18 @@ class RateLimitingQueue:
                 item = self._queue.popleft()
                 self._dirty.discard(item)
                 self._processing.add(item)
+                if self._queue:
+                    # chained wakeup: the delayed drain readied more items
+                    # than this getter consumes
+                    self._wake_one()
+                self._rearm_timer()
                 return item
             if self._shutdown:
                 return None
-            timeout = self._next_delay()
+            self._rearm_timer()
             fut = loop.create_future()
             fut._parked = True  # type: ignore[attr-defined]
             self._getters.append(fut)
-            handle = None
-            if timeout is not None:
-                # timer resolves the SAME future: no lost-wakeup window, and
-                # a timer-woken getter just recomputes the next delay
-                handle = loop.call_later(
-                    timeout, lambda f=fut: None if f.done() else f.set_result(None)
-                )
             try:
                 await fut
             except asyncio.CancelledError:
@@ -173,11 +213,9 @@ class RateLimitingQueue:
                     self._wake_one()
                 raise
             finally:
-                if handle is not None:
-                    handle.cancel()
                 # O(1) in the common case: _wake_one/_wake_all already
-                # unparked us; only a timer/cancel exit still sits in the
-                # deque (a full-deque remove per get was O(workers) each)
+                # unparked us; only a cancel exit still sits in the deque
+                # (a full-deque remove per get was O(workers) each)
                 if getattr(fut, "_parked", False):
                     try:
                         self._getters.remove(fut)
@@ -198,6 +236,9 @@ class RateLimitingQueue:
 
     async def shutdown(self) -> None:
         self._shutdown = True
+        if self._timer is not None:
+            self._timer.cancel()
+            self._timer = None
         self._wake_all()
 
     # -- helpers ------------------------------------------------------------
