@@ -125,3 +125,60 @@ def test_delayed_add_timer_wakes_parked_getter():
         assert item == "later"
 
     run(main())
+
+
+def test_shared_timer_rearms_to_earlier_deadline():
+    """A delayed add with a SOONER deadline than the armed shared timer
+    must re-arm it — otherwise the earlier item waits for the later
+    deadline (lost-wakeup class introduced by the one-timer design)."""
+
+    async def main():
+        import time
+
+        q = RateLimitingQueue()
+        worker_got = []
+
+        async def worker():
+            item = await q.get()
+            worker_got.append((item, time.monotonic()))
+
+        t = asyncio.create_task(worker())
+        await asyncio.sleep(0.01)  # park the getter (no items)
+        await q.add_after("late", 5.0)   # timer armed at +5 s
+        await q.add_after("soon", 0.05)  # must re-arm to +0.05 s
+        t0 = time.monotonic()
+        await asyncio.wait_for(t, 2.0)
+        item, when = worker_got[0]
+        assert item == "soon"
+        assert when - t0 < 1.0, f"woke after {when - t0:.2f}s — timer not re-armed"
+        await q.shutdown()
+
+    run(main())
+
+
+def test_shared_timer_chained_wakeups_drain_batch():
+    """When one timer tick readies MANY delayed items, chained wakeups must
+    hand one to each parked worker (no single-consumer bottleneck)."""
+
+    async def main():
+        q = RateLimitingQueue()
+        got = []
+
+        async def worker():
+            while True:
+                item = await q.get()
+                if item is None:
+                    return
+                got.append(item)
+                await q.done(item)
+
+        workers = [asyncio.create_task(worker()) for _ in range(8)]
+        await asyncio.sleep(0.01)
+        for i in range(50):
+            await q.add_after(f"i{i}", 0.05)  # all due at the same instant
+        await asyncio.sleep(0.5)
+        assert sorted(got) == sorted(f"i{i}" for i in range(50)), got
+        await q.shutdown()
+        await asyncio.gather(*workers)
+
+    run(main())
