@@ -126,13 +126,21 @@ class Controller:
                 continue
             self._m_dur.observe(time.monotonic() - start)
             self._m_ok.inc()
-            self.queue.forget(key)
-            await self.queue.done(key)
-            if result is not None:
-                if result.requeue_after is not None:
-                    await self.queue.add_after(key, result.requeue_after)
-                elif result.requeue:
-                    await self.queue.add_rate_limited(key)
+            # controller-runtime semantics: Requeue=true retries RETAIN the
+            # rate-limiter state so contention retries back off
+            # exponentially; RequeueAfter and clean completion reset it
+            # (forgetting before every success kept conflict storms at the
+            # 5 ms base forever)
+            if result is not None and result.requeue_after is not None:
+                self.queue.forget(key)
+                await self.queue.done(key)
+                await self.queue.add_after(key, result.requeue_after)
+            elif result is not None and result.requeue:
+                await self.queue.done(key)
+                await self.queue.add_rate_limited(key)
+            else:
+                self.queue.forget(key)
+                await self.queue.done(key)
 
 
 class SingletonController(Controller):
