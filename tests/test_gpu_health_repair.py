@@ -114,3 +114,50 @@ def test_gpu_recovery_within_toleration_cancels_repair():
             await h.stop()
 
     run(main())
+
+
+def test_patch_node_condition_retries_conflicts_with_fresh_read():
+    """the status patch carries the read's
+    resourceVersion; on a conflicting concurrent write it re-reads and
+    retries, so a kubelet condition written in between is never clobbered
+    or resurrected."""
+
+    async def main():
+        from gpu_provisioner_amd.kube.client import ConflictError
+
+        h = Harness().add_all_controllers(gc_interval=60.0, with_health=False)
+        await h.start()
+        try:
+            await h.kube.create(h.make_nodeclaim("cond2"))
+            nc = await h.wait_initialized("cond2")
+            node_name = nc["status"]["nodeName"]
+
+            # kubelet writes a condition the agent's earlier read has not
+            # seen; the agent's FIRST patch attempt conflicts (injected),
+            # forcing the retry path against a fresh read
+            node = await h.kube.get("v1", "Node", node_name)
+            ko.set_condition(node, "KubeletFresh", "True", "Heartbeat", "")
+            await h.kube.update_status(node)
+
+            fired = {"n": 0}
+
+            def conflict_once(verb, gvk, payload):
+                if verb == "patch" and gvk == ("v1", "Node") and fired["n"] == 0:
+                    fired["n"] += 1
+                    return ConflictError("injected concurrent write")
+                return None
+
+            h.server.reactors.append(conflict_once)
+            await patch_node_condition(h.kube, node_name, healthy_report())
+            h.server.reactors.clear()
+            assert fired["n"] == 1  # the retry actually happened
+
+            after = await h.kube.get("v1", "Node", node_name)
+            assert ko.condition_is_true(after, karpv1.AMD_GPU_HEALTHY_CONDITION_TYPE)
+            # the kubelet's concurrent condition survived the merge
+            assert ko.condition_is_true(after, "KubeletFresh")
+            assert ko.node_is_ready(after)
+        finally:
+            await h.stop()
+
+    run(main())
