@@ -247,3 +247,29 @@ def test_shipped_sku_override_example_loads():
     assert cat.get("Standard_ND96isr_MI400X_v7") is not None
     assert cat.gpu_count("Standard_ND96isr_MI400X_v7") == 8
     assert cat.get("Standard_ND32is_MI355X_v6") is None
+
+
+def test_event_recorder_dedupe_window_capped():
+    """The dedupe window is bounded (r02): high-churn fleets key entries by
+    object uid, and an uncapped rate x TTL window measured tens of MB."""
+    from gpu_provisioner_amd.events import recorder as rec_mod
+
+    async def main():
+        server = InMemoryAPIServer()
+        kube = InMemoryClient(server)
+        r = rec_mod.EventRecorder(kube)
+        for i in range(rec_mod.DEDUPE_MAX_ENTRIES + 500):
+            obj = {"apiVersion": "v1", "kind": "Node",
+                   "metadata": {"name": f"n{i}", "uid": f"u{i}"}}
+            r.publish(obj, "Reason", "msg")
+        assert len(r._seen) == rec_mod.DEDUPE_MAX_ENTRIES
+        # dedupe still works for a key inside the window
+        before = len(r._pending)
+        r.publish({"apiVersion": "v1", "kind": "Node",
+                   "metadata": {"name": "nlast", "uid": f"u{rec_mod.DEDUPE_MAX_ENTRIES+499}"}},
+                  "Reason", "msg")
+        # same (uid, reason, type) as the most recent publish → deduped
+        assert len(r._seen) == rec_mod.DEDUPE_MAX_ENTRIES
+        await asyncio.gather(*list(r._pending), return_exceptions=True)
+
+    run(main())
